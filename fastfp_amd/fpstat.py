@@ -1,0 +1,90 @@
+"""Plain Fp-statistic API: :class:`FastFp`.
+
+API parity with the reference's ``FastFp``
+(``/root/reference/fastfp/fastfp.py:22-101``): construction from
+``(psrs, pta)``, a per-frequency ``calculate_Fp(fgw, Nvecs, Ts, sigmas)``
+and ``__call__``.  Additionally provides the production entry point
+:meth:`sweep` which evaluates the whole frequency grid through the
+restructured :class:`fastfp_amd.engine.FpEngine` (one Cholesky per
+pulsar, one GEMM per pulsar — not 6 Woodbury solves per (f, pulsar)).
+"""
+
+from __future__ import annotations
+
+import math
+
+import numpy as np
+import torch
+
+from fastfp_amd.engine import FpEngine
+from fastfp_amd.xcy import get_xCy
+
+
+class FastFp:
+    """Fp detection statistic (Ellis, Siemens & Creighton 2012)."""
+
+    def __init__(self, psrs, pta=None):
+        self.psrs = psrs
+        self.pta = pta
+        self.toas = [np.asarray(p.toas, dtype=np.float64) for p in psrs]
+        self.residuals = [np.asarray(p.residuals, dtype=np.float64) for p in psrs]
+
+    def __call__(self, fgw, Nvecs, Ts, sigmas):
+        return self.calculate_Fp(fgw, Nvecs, Ts, sigmas)
+
+    # ------------------------------------------------------------------
+    # parity path: single frequency, explicit Woodbury products
+    # ------------------------------------------------------------------
+    def calculate_Fp(self, fgw, Nvecs, Ts, sigmas) -> float:
+        """Single-frequency Fp — same evaluation order as the reference
+        (filter amplitude ``fgw^-1/3`` included for bit-level parity,
+        ``/root/reference/fastfp/fastfp.py:69-92``)."""
+        fstat = 0.0
+        amp = 1.0 / fgw ** (1.0 / 3.0)
+        for Nvec, T, sigma, toa, resid in zip(
+            Nvecs, Ts, sigmas, self.toas, self.residuals
+        ):
+            arg = 2.0 * math.pi * fgw * toa
+            A0 = amp * np.sin(arg)
+            A1 = amp * np.cos(arg)
+
+            ip1 = get_xCy(Nvec, T, sigma, A0, resid)
+            ip2 = get_xCy(Nvec, T, sigma, A1, resid)
+            N = np.array([ip1, ip2])
+
+            M = np.empty((2, 2))
+            M[0, 0] = get_xCy(Nvec, T, sigma, A0, A0)
+            M[0, 1] = get_xCy(Nvec, T, sigma, A0, A1)
+            M[1, 0] = M[0, 1]
+            M[1, 1] = get_xCy(Nvec, T, sigma, A1, A1)
+
+            fstat += 0.5 * float(N @ np.linalg.solve(M, N))
+        return fstat
+
+    # ------------------------------------------------------------------
+    # production path: whole grid at once on the engine
+    # ------------------------------------------------------------------
+    def sweep(
+        self,
+        freqs,
+        Nvecs,
+        Ts,
+        sigmas,
+        device: str = None,
+        freq_chunk: int = 2048,
+    ) -> np.ndarray:
+        """Fp over a frequency grid.  Returns (F,) numpy array."""
+        if device is None:
+            device = "cuda" if torch.cuda.is_available() else "cpu"
+        eng = FpEngine(self.psrs, Nvecs, Ts, device=device)
+        eng.precompute(freqs, freq_chunk=freq_chunk)
+        fp = eng.sweep(sigmas=sigmas)
+        return fp.cpu().numpy()
+
+
+def compute_Fp(psrs, pta, noise, freqs, device=None) -> np.ndarray:
+    """One-call convenience: precompute mats and sweep the grid."""
+    from fastfp_amd.model import get_mats_fp
+
+    Nvecs, Ts, sigmas = get_mats_fp(pta, noise)
+    return FastFp(psrs, pta).sweep(freqs, Nvecs, Ts, sigmas, device=device)

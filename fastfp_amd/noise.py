@@ -1,0 +1,316 @@
+"""Noise models: white-noise variances and the phi(theta) prior engine
+(per-pulsar red noise, common uncorrelated red noise, GP-ECORR).
+
+This is the native re-design of the reference's container layer
+(``RN_container``/``CURN_container``/``GPEcorr_container``,
+``/root/reference/fastfp/nmfp.py:131-477``).  Differences by design:
+
+- containers are **draw-vectorized**: parameter-dict values may be
+  scalars or length-D arrays, and ``get_phiinv`` returns shape ``(m,)``
+  or ``(D, m)`` torch float64 tensors (the reference vectorizes with an
+  outer ``jax.vmap`` instead, ``/root/reference/examples/run_nmfp.py:265-266``),
+- phi lives on whatever torch device the engine runs on.
+
+Conventions preserved exactly (validated by tests against the dense
+oracle):
+
+- power-law PSD ``phi = f^-gamma A^2 /(12 pi^2) fyr^(gamma-3) * df``
+  with ``df`` the diff of the unique frequencies, repeated pairwise
+  (``/root/reference/fastfp/nmfp.py:217-234``),
+- CURN phi **added** onto the first ``2*ngwb`` red-noise bins
+  (``/root/reference/fastfp/nmfp.py:247``),
+- block order tm | ecorr | rn(+curn) (``/root/reference/fastfp/nmfp.py:282``),
+- timing-model prior 1e40 * ones (``/root/reference/fastfp/nmfp.py:267``),
+- phi is diagonal throughout; ``phiinv = 1/phi``.
+"""
+
+from __future__ import annotations
+
+import numpy as np
+import torch
+
+from fastfp_amd.constants import fyr
+from fastfp_amd.bases import create_freqarray, ecorr_basis_by_backend
+
+TM_PRIOR = 1e40
+
+
+def _as_tensor(x, device=None):
+    if isinstance(x, torch.Tensor):
+        t = x.to(dtype=torch.float64)
+    else:
+        t = torch.as_tensor(np.asarray(x), dtype=torch.float64)
+    if device is not None:
+        t = t.to(device)
+    return t
+
+
+def white_noise_nvec(psr, noise: dict = None, simple_wn: bool = True,
+                     select: str = "backend") -> np.ndarray:
+    """Diagonal white-noise variances N per TOA.
+
+    ``simple_wn``: EFAC = 1.0 (the reference's simulated-data default,
+    ``/root/reference/fastfp/utils.py:147-149``): N = toaerr^2.
+
+    Otherwise per-backend EFAC/EQUAD with the TEMPO2 (t2equad)
+    convention used by enterprise's ``white_noise_block``:
+    ``N = efac^2 * (toaerr^2 + t2equad^2)`` with parameters read from the
+    noise dict as ``{psr}_{backend}_efac`` / ``{psr}_{backend}_log10_t2equad``
+    (missing keys default to efac=1, equad=0).
+    """
+    if simple_wn:
+        return psr.toaerrs**2
+    nvec = np.zeros(psr.ntoa, dtype=np.float64)
+    keysel = (
+        np.unique(psr.backend_flags)
+        if select == "backend"
+        else np.array(["all"], dtype=object)
+    )
+    for b in keysel:
+        mask = (
+            np.asarray(psr.backend_flags == b)
+            if select == "backend"
+            else np.ones(psr.ntoa, dtype=bool)
+        )
+        efac = float(noise.get(f"{psr.name}_{b}_efac", 1.0)) if noise else 1.0
+        l10eq = noise.get(f"{psr.name}_{b}_log10_t2equad", None) if noise else None
+        equad2 = 10.0 ** (2.0 * float(l10eq)) if l10eq is not None else 0.0
+        nvec[mask] = efac**2 * (psr.toaerrs[mask] ** 2 + equad2)
+    return nvec
+
+
+def powerlaw_psd(Ffreqs, log10_A, gamma) -> np.ndarray:
+    """NumPy scalar-parameter power-law PSD (used by the synthetic data
+    generator).  Same formula as :meth:`RNContainer._powerlaw`."""
+    Ffreqs = np.asarray(Ffreqs, dtype=np.float64)
+    funique = Ffreqs[::2]
+    df = np.diff(np.concatenate(([0.0], funique)))
+    return (
+        Ffreqs ** (-gamma)
+        * (10.0**log10_A) ** 2
+        / 12.0
+        / np.pi**2
+        * fyr ** (gamma - 3.0)
+        * np.repeat(df, 2)
+    )
+
+
+def _powerlaw_torch(Ffreqs: torch.Tensor, log10_A: torch.Tensor,
+                    gamma: torch.Tensor) -> torch.Tensor:
+    """Draw-vectorized power law: log10_A/gamma scalars or (D,), returns
+    (nf,) or (D, nf)."""
+    funique = Ffreqs[::2]
+    df = torch.diff(
+        torch.cat([torch.zeros(1, dtype=Ffreqs.dtype, device=Ffreqs.device), funique])
+    )
+    dff = torch.repeat_interleave(df, 2)
+    log10_A = torch.atleast_1d(log10_A)
+    gamma = torch.atleast_1d(gamma)
+    # (D,1) exponents against (nf,) freqs
+    phi = (
+        Ffreqs[None, :] ** (-gamma[:, None])
+        * (10.0 ** log10_A[:, None]) ** 2
+        / 12.0
+        / np.pi**2
+        * fyr ** (gamma[:, None] - 3.0)
+        * dff[None, :]
+    )
+    return phi.squeeze(0) if phi.shape[0] == 1 else phi
+
+
+class CURNContainer:
+    """Common uncorrelated red-noise (GWB proxy) phi engine.
+
+    Parity with ``CURN_container`` (``/root/reference/fastfp/nmfp.py:344-417``):
+    fixed parameter names ``gw_log10_A`` / ``gw_gamma``.
+    """
+
+    def __init__(self, Ffreqs, device=None):
+        self.rn_A_name = "gw_log10_A"
+        self.rn_gam_name = "gw_gamma"
+        self.Ffreqs = _as_tensor(Ffreqs, device)
+
+    def to(self, device):
+        self.Ffreqs = self.Ffreqs.to(device)
+        return self
+
+    def _powerlaw(self, pars: dict) -> torch.Tensor:
+        return _powerlaw_torch(
+            self.Ffreqs,
+            _as_tensor(pars[self.rn_A_name], self.Ffreqs.device),
+            _as_tensor(pars[self.rn_gam_name], self.Ffreqs.device),
+        )
+
+    def get_phi_curn(self, pars: dict) -> torch.Tensor:
+        return self._powerlaw(pars)
+
+    def update_phi(self, pars: dict) -> torch.Tensor:
+        return self.get_phi_curn(pars)
+
+    def get_phiinv(self, pars: dict) -> torch.Tensor:
+        return 1.0 / self.update_phi(pars)
+
+
+class GPEcorrContainer:
+    """Per-backend GP-ECORR phi engine (fixed, precomputed).
+
+    Parity with ``GPEcorr_container`` (``/root/reference/fastfp/nmfp.py:420-477``):
+    reads ``{psr}_basis_ecorr_{backend}_log10_ecorr`` from the noise dict
+    (backends in sorted/np.unique order) and precomputes the fixed phi
+    slice ``weights_b * 10^(2 log10_ecorr_b)`` per backend.
+    """
+
+    def __init__(self, psr, weights: list = None, fix_wn_vals: dict = None,
+                 device=None):
+        self.psr = psr
+        if weights is None:
+            _, weights = ecorr_basis_by_backend(psr)
+        self.weights = weights
+        self.fix_wn_vals = fix_wn_vals or {}
+
+        backends = np.unique(psr.backend_flags)
+        self.ecorrs = np.array(
+            [
+                float(
+                    self.fix_wn_vals[
+                        "_".join([psr.name, "basis", "ecorr", str(b), "log10_ecorr"])
+                    ]
+                )
+                for b in backends
+            ],
+            dtype=np.float64,
+        )
+        slcs = [
+            np.asarray(w, dtype=np.float64) * 10.0 ** (2.0 * e)
+            for w, e in zip(self.weights, self.ecorrs)
+        ]
+        phi = (
+            np.concatenate(slcs)
+            if slcs
+            else np.zeros(0, dtype=np.float64)
+        )
+        self._phi = _as_tensor(phi, device)
+
+    def to(self, device):
+        self._phi = self._phi.to(device)
+        return self
+
+    def get_phi(self, pars: dict = None) -> torch.Tensor:
+        return self._phi
+
+
+class RNContainer:
+    """Per-pulsar red-noise phi engine.
+
+    Re-design of ``RN_container`` (``/root/reference/fastfp/nmfp.py:131-341``).
+    Instead of the reference's 8 hand-written phi-variant methods chosen at
+    init (``:188-199``), one vectorized assembly handles every
+    {tm, ecorr, curn} presence combination; dedicated ``get_phi_*``
+    methods are kept as thin wrappers for API parity.
+    """
+
+    def __init__(
+        self,
+        psr,
+        Ffreqs=None,
+        ncomps: int = 30,
+        gp_ecorr: bool = False,
+        ecorr_container: GPEcorrContainer = None,
+        add_curn: bool = False,
+        curn_container: CURNContainer = None,
+        inc_tm: bool = True,
+        device=None,
+    ):
+        self.psr = psr
+        self.ncomps = ncomps
+        self.gp_ecorr = gp_ecorr
+        self.ecorr_container = ecorr_container
+        self.add_curn = add_curn
+        self.curn_container = curn_container
+        self.inc_tm = inc_tm
+
+        self.rn_A_name = f"{psr.name}_red_noise_log10_A"
+        self.rn_gam_name = f"{psr.name}_red_noise_gamma"
+
+        if Ffreqs is None:
+            Ffreqs = create_freqarray(psr.Tspan, ncomps)
+        self.Ffreqs = _as_tensor(Ffreqs, device)
+        self.tm_weights = (
+            torch.ones(psr.ntm, dtype=torch.float64, device=self.Ffreqs.device)
+            if inc_tm
+            else torch.zeros(0, dtype=torch.float64, device=self.Ffreqs.device)
+        )
+
+    def to(self, device):
+        self.Ffreqs = self.Ffreqs.to(device)
+        self.tm_weights = self.tm_weights.to(device)
+        if self.ecorr_container is not None:
+            self.ecorr_container.to(device)
+        if self.curn_container is not None:
+            self.curn_container.to(device)
+        return self
+
+    # ------------------------------------------------------------------
+    def _powerlaw(self, pars: dict) -> torch.Tensor:
+        return _powerlaw_torch(
+            self.Ffreqs,
+            _as_tensor(pars[self.rn_A_name], self.Ffreqs.device),
+            _as_tensor(pars[self.rn_gam_name], self.Ffreqs.device),
+        )
+
+    def update_phi(self, pars: dict) -> torch.Tensor:
+        """Assemble diagonal phi: [tm | ecorr | rn(+curn)], shape (m_phi,)
+        for scalar pars or (D, m_phi) for length-D pars."""
+        rn_phi = self._powerlaw(pars)  # (nf,) or (D, nf)
+        if self.add_curn:
+            curn_phi = self.curn_container.get_phi_curn(pars)
+            nc = curn_phi.shape[-1]
+            rn_phi = rn_phi.clone()
+            rn_phi[..., :nc] = rn_phi[..., :nc] + curn_phi
+        blocks = []
+        batched = rn_phi.dim() == 2
+        D = rn_phi.shape[0] if batched else None
+
+        def expand(vec):
+            if batched:
+                return vec[None, :].expand(D, -1)
+            return vec
+
+        if self.inc_tm:
+            blocks.append(expand(self.tm_weights * TM_PRIOR))
+        if self.gp_ecorr:
+            blocks.append(expand(self.ecorr_container.get_phi(pars)))
+        blocks.append(rn_phi)
+        return torch.cat(blocks, dim=-1)
+
+    def get_phiinv(self, pars: dict) -> torch.Tensor:
+        return 1.0 / self.update_phi(pars)
+
+    # --- API-parity thin wrappers (reference's 8 variants, :239-292) ---
+    def get_phi_rn(self, pars):
+        return self._powerlaw(pars)
+
+    def get_phi_tm_rn(self, pars):
+        assert self.inc_tm and not self.gp_ecorr and not self.add_curn
+        return self.update_phi(pars)
+
+    def get_phi_tm_rn_curn(self, pars):
+        assert self.inc_tm and not self.gp_ecorr and self.add_curn
+        return self.update_phi(pars)
+
+    def get_phi_tm_ecorr_rn(self, pars):
+        assert self.inc_tm and self.gp_ecorr and not self.add_curn
+        return self.update_phi(pars)
+
+    def get_phi_tm_ecorr_rn_curn(self, pars):
+        assert self.inc_tm and self.gp_ecorr and self.add_curn
+        return self.update_phi(pars)
+
+    @property
+    def nphi(self) -> int:
+        n = self.Ffreqs.shape[0]
+        if self.inc_tm:
+            n += self.tm_weights.shape[0]
+        if self.gp_ecorr:
+            n += self.ecorr_container.get_phi().shape[0]
+        return n

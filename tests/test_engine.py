@@ -1,0 +1,156 @@
+"""Engine correctness: the restructured sweep vs the dense oracle and vs
+the reference-shaped parity path."""
+
+import numpy as np
+import pytest
+import torch
+
+from fastfp_amd import (
+    FastFp,
+    FpEngine,
+    NMFp,
+    get_mats_fp,
+    get_mats_nmfp,
+    initialize_pta,
+    make_synthetic_pta,
+)
+from fastfp_amd.bases import create_freqarray, fourier_basis, timing_model_basis_svd
+from fastfp_amd.noise import white_noise_nvec
+from oracle import dense_fp_sweep
+
+
+def _tiny_setup(seed=0, npsr=3, ntoa=70, ncomps=4, ntm=3, tm_prior=1e5):
+    """Small PTA with MODERATE priors so the dense oracle is well
+    conditioned."""
+    psrs = make_synthetic_pta(
+        npsr=npsr, ntoa=ntoa, tspan_yr=10.0, ntm=ntm, seed=seed, toaerr=1e-6
+    )
+    rng = np.random.default_rng(seed + 100)
+    Nvecs, Ts, phis = [], [], []
+    for p in psrs:
+        Nvecs.append(white_noise_nvec(p))
+        U = timing_model_basis_svd(p.Mmat)
+        F = fourier_basis(p.toas, create_freqarray(p.Tspan, ncomps))
+        T = np.concatenate([U, F], axis=1)
+        Ts.append(T)
+        # phi: tm block (moderate), rn block random positive
+        phi = np.concatenate(
+            [
+                np.full(U.shape[1], tm_prior) * 1e-12,  # tm, scaled to residual^2
+                rng.uniform(0.5, 2.0, F.shape[1]) * 1e-12,
+            ]
+        )
+        phis.append(phi)
+    return psrs, Nvecs, Ts, phis
+
+
+def test_engine_fp_matches_dense_oracle():
+    psrs, Nvecs, Ts, phis = _tiny_setup()
+    freqs = np.linspace(3e-9, 5e-8, 7)
+    want = dense_fp_sweep(psrs, Nvecs, Ts, phis, freqs)
+
+    eng = FpEngine(psrs, Nvecs, Ts, device="cpu")
+    eng.precompute(freqs, freq_chunk=3)
+    phiinvs = [1.0 / p for p in phis]
+    got = eng.sweep(phiinvs=phiinvs).numpy()
+    np.testing.assert_allclose(got, want, rtol=1e-7)
+
+
+def test_engine_sigma_and_phiinv_paths_agree():
+    psrs, Nvecs, Ts, phis = _tiny_setup(seed=1)
+    freqs = np.linspace(3e-9, 5e-8, 5)
+    eng = FpEngine(psrs, Nvecs, Ts, device="cpu")
+    eng.precompute(freqs)
+    phiinvs = [1.0 / p for p in phis]
+    sigmas = [
+        T.T @ (T / nv[:, None]) + np.diag(pi)
+        for T, nv, pi in zip(Ts, Nvecs, phiinvs)
+    ]
+    a = eng.sweep(phiinvs=phiinvs).numpy()
+    b = eng.sweep(sigmas=sigmas).numpy()
+    np.testing.assert_allclose(a, b, rtol=1e-10)
+
+
+def test_parity_calculate_fp_matches_engine():
+    """The reference-shaped get_xCy path (with the f^-1/3 amplitude)
+    equals the engine sweep (amplitude dropped) — proves the amplitude
+    cancellation."""
+    psrs, Nvecs, Ts, phis = _tiny_setup(seed=2)
+    freqs = np.linspace(4e-9, 4e-8, 4)
+    phiinvs = [1.0 / p for p in phis]
+    sigmas = [
+        T.T @ (T / nv[:, None]) + np.diag(pi)
+        for T, nv, pi in zip(Ts, Nvecs, phiinvs)
+    ]
+    fp_obj = FastFp(psrs)
+    want = np.array(
+        [fp_obj.calculate_Fp(f, Nvecs, Ts, sigmas) for f in freqs]
+    )
+    got = fp_obj.sweep(freqs, Nvecs, Ts, sigmas, device="cpu")
+    np.testing.assert_allclose(got, want, rtol=1e-7)
+
+
+def test_fp_invariant_to_tm_prior_scale():
+    """1e40 improper-flat tm prior acts as a projection: Fp must be
+    insensitive to the exact (huge) scale."""
+    psrs, Nvecs, Ts, phis = _tiny_setup(seed=3)
+    freqs = np.linspace(4e-9, 4e-8, 3)
+    eng = FpEngine(psrs, Nvecs, Ts, device="cpu")
+    eng.precompute(freqs)
+    ntm = 3
+    out = []
+    for scale in (1e30, 1e40):
+        phiinvs = []
+        for phi in phis:
+            p = phi.copy()
+            p[:ntm] = scale
+            phiinvs.append(1.0 / p)
+        out.append(eng.sweep(phiinvs=phiinvs).numpy())
+    np.testing.assert_allclose(out[0], out[1], rtol=1e-5)
+
+
+def test_nmfp_sweep_matches_per_draw_parity():
+    """Draw-vectorized NM-Fp sweep == per-draw calculate_nmfp (which goes
+    through get_xCy), on a full model with tm(1e40)+rn+curn."""
+    psrs = make_synthetic_pta(npsr=2, ntoa=60, tspan_yr=10.0, ntm=3, seed=4)
+    noise = {"gw_gamma": 13.0 / 3.0, "gw_log10_A": np.log10(2e-15)}
+    pta = initialize_pta(psrs, noise, inc_cp=True, rn_comps=4, gwb_comps=3)
+    TNTs, Nvecs, Ts = get_mats_nmfp(pta, noise)
+
+    D = 3
+    rng = np.random.default_rng(5)
+    samples = {}
+    for name in pta.params:
+        if name.endswith("gamma"):
+            samples[name] = rng.uniform(2.0, 6.0, D)
+        else:
+            samples[name] = rng.uniform(-15.0, -13.0, D)
+
+    nm = NMFp(psrs, pta.rn_containers)
+    freqs = np.linspace(4e-9, 4e-8, 4)
+    got = nm.sweep(freqs, samples, Nvecs, Ts, device="cpu", draw_chunk=2)
+    assert got.shape == (D, len(freqs))
+
+    for d in range(D):
+        pars = {k: float(v[d]) for k, v in samples.items()}
+        for j, f in enumerate(freqs):
+            want = nm.calculate_nmfp(f, pars, Nvecs, Ts, TNTs)
+            assert got[d, j] == pytest.approx(want, rel=1e-6)
+
+
+def test_chi2_null_distribution():
+    """Statistical integration test: for noise-only
+    data 2*Fp ~ chi^2(2 * npsr); check the sample mean loosely."""
+    npsr = 6
+    psrs = make_synthetic_pta(npsr=npsr, ntoa=150, tspan_yr=12.0, ntm=3, seed=6)
+    noise = {}
+    pta = initialize_pta(psrs, noise, inc_cp=False, rn_comps=5)
+    Nvecs, Ts, sigmas = get_mats_fp(
+        pta, {f"{p.name}_red_noise_gamma": 4.0 for p in psrs}
+        | {f"{p.name}_red_noise_log10_A": -16.0 for p in psrs}
+    )
+    freqs = np.linspace(5e-9, 8e-8, 40)
+    fp = FastFp(psrs).sweep(freqs, Nvecs, Ts, sigmas, device="cpu")
+    mean2fp = float(np.mean(2.0 * fp))
+    expect = 2.0 * npsr
+    assert abs(mean2fp - expect) < 0.45 * expect, (mean2fp, expect)
