@@ -1,0 +1,171 @@
+"""fastfp_amd flagship benchmark — the BASELINE.json headline metric.
+
+Metric: **Fp evals/sec** (= frequencies x draws evaluated per second) on
+the 67-pulsar noise-marginalized Fp sweep (BASELINE.json config 3:
+1e4 draws x 1e3 freqs at 8 GPUs).  One "step" = one draw batch of
+``--draws-per-step`` MCMC noise draws evaluated at all ``--freqs`` CW
+frequencies across all pulsars (phi(theta) assembly + Sigma assembly +
+batched Cholesky + fused triangular-solve/2x2-reduction, all on device).
+Weak scaling: each rank owns its own draw batch; the final RCCL
+all-gather of the per-rank (D_local, F) spectrum shards is included in
+the timed region.
+
+Reference baseline: ~6.0 Fp evals/s (200 freqs / 33.56 s, 1 draw,
+unknown hardware — BASELINE.md); vs_baseline = value / 6.0.
+
+Launch (driver contract):
+  python bench.py --gpus 1 --steps K --warmup W
+  python -m torch.distributed.run --nnodes=1 --nproc-per-node N \
+      --master-addr 127.0.0.1 bench.py --gpus N --steps K --warmup W
+
+Data: synthetic epoch-structured PTA (random-init) of the BASELINE
+config shape — there is no network for real datasets.
+"""
+
+import argparse
+import json
+import os
+import time
+
+import numpy as np
+import torch
+
+from fastfp_amd.data import make_synthetic_pta
+from fastfp_amd.engine import FpEngine
+from fastfp_amd.model import get_mats_nmfp, initialize_pta
+from fastfp_amd.parallel import all_gather_concat, cleanup, init_distributed
+
+
+def build_problem(args, device, rank):
+    psrs = make_synthetic_pta(
+        npsr=args.npsr,
+        ntoa=args.ntoa,
+        tspan_yr=15.0,
+        ntm=args.ntm,
+        seed=1234,  # same PTA on every rank
+        ragged=True,
+    )
+    noise = {"gw_gamma": 13.0 / 3.0, "gw_log10_A": float(np.log10(2e-15))}
+    for p in psrs:
+        noise[f"{p.name}_red_noise_gamma"] = 13.0 / 3.0
+        noise[f"{p.name}_red_noise_log10_A"] = -14.5
+    pta = initialize_pta(
+        psrs, noise, inc_cp=True, rn_comps=args.rn_comps, gwb_comps=args.gwb_comps
+    )
+    TNTs, Nvecs, Ts = get_mats_nmfp(pta, noise)
+
+    eng = FpEngine(psrs, Nvecs, Ts, device=device)
+    freqs = np.arange(1, args.freqs + 1) / pta.Tspan
+    eng.precompute(freqs, freq_chunk=args.freq_chunk)
+
+    for cont in pta.rn_containers:
+        cont.to(device)
+
+    # pre-generate a pool of noise-parameter draws on device (rank-seeded:
+    # each rank marginalizes over its own draws -- weak scaling)
+    rng = np.random.default_rng(1000 + rank)
+    D = args.draws_per_step
+    pool = {}
+    for name in pta.params:
+        if name.endswith("gamma"):
+            v = rng.uniform(1.0, 6.5, D)
+        else:
+            v = rng.uniform(-16.0, -13.5, D)
+        pool[name] = torch.as_tensor(v, dtype=torch.float64, device=device)
+    return pta, eng, pool
+
+
+def run_step(pta, eng, pool, args, fp_accum):
+    phiinvs = [cont.get_phiinv(pool) for cont in pta.rn_containers]
+    fp_accum.zero_()
+    eng.sweep(phiinvs=phiinvs, draw_chunk=args.draw_chunk, accumulate_to=fp_accum)
+    return fp_accum
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=5)
+    ap.add_argument("--warmup", type=int, default=2)
+    ap.add_argument("--npsr", type=int, default=67)
+    ap.add_argument("--ntoa", type=int, default=5000)
+    ap.add_argument("--ntm", type=int, default=60)
+    ap.add_argument("--rn-comps", type=int, default=30)
+    ap.add_argument("--gwb-comps", type=int, default=30)
+    ap.add_argument("--freqs", type=int, default=1000)
+    ap.add_argument("--draws-per-step", type=int, default=500)
+    ap.add_argument("--draw-chunk", type=int, default=64)
+    ap.add_argument("--freq-chunk", type=int, default=4096)
+    ap.add_argument("--device", type=str, default=None)
+    args = ap.parse_args()
+
+    rank, world, device = init_distributed(
+        device=torch.device(args.device) if args.device else None
+    )
+    on_gpu = device.type == "cuda"
+
+    pta, eng, pool = build_problem(args, device, rank)
+    F = args.freqs
+    D = args.draws_per_step
+    fp_accum = torch.zeros((D, F), dtype=torch.float64, device=device)
+
+    def barrier_sync():
+        if world > 1:
+            torch.distributed.barrier()
+        if on_gpu:
+            torch.cuda.synchronize()
+
+    for _ in range(args.warmup):
+        run_step(pta, eng, pool, args, fp_accum)
+
+    barrier_sync()
+    t0 = time.perf_counter()
+    for _ in range(args.steps):
+        run_step(pta, eng, pool, args, fp_accum)
+    # the spectrum all-gather is part of the job
+    full = all_gather_concat(fp_accum, world, dim=0)
+    barrier_sync()
+    elapsed = time.perf_counter() - t0
+
+    # max over ranks
+    t = torch.tensor([elapsed], dtype=torch.float64, device=device if on_gpu else "cpu")
+    if world > 1:
+        torch.distributed.all_reduce(t, op=torch.distributed.ReduceOp.MAX)
+    elapsed = float(t.item())
+
+    evals = args.steps * D * F * world  # whole-job evals
+    value = evals / elapsed
+    ms_per_step = elapsed / args.steps * 1000.0
+
+    if rank == 0:
+        out = {
+            "metric": "Fp evals/sec (freqs*draws/s), 67-psr NMFp sweep",
+            "value": value,
+            "unit": "evals/s",
+            "n_gpus": world,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": ms_per_step,
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": value / 6.0,
+            "dtype": "fp64",
+            "data": "synthetic",
+            "config": {
+                "model": "nmfp",
+                "npsr": args.npsr,
+                "ntoa": args.ntoa,
+                "basis_size": args.ntm + 2 * args.rn_comps,
+                "freqs": F,
+                "draws_per_step_per_gpu": D,
+                "global_batch": D * world,
+                "parallelism": f"dp{world} draw-sharded",
+                "spectrum_shape": list(full.shape),
+            },
+        }
+        print(json.dumps(out))
+    cleanup()
+
+
+if __name__ == "__main__":
+    main()

@@ -1,0 +1,199 @@
+"""Noise-marginalized Fp CLI.
+
+Argument and output parity with the reference script
+(``/root/reference/examples/run_nmfp.py``): positionals
+(psrfile, noisefile, chainfile, savefile), flags ``--inc_ecorr
+--inc_cp --nrncomps --ngwbcomps --ncwfreqs --nsamples --batch_size``,
+chain handling (25% burn-in, last 4 bookkeeping columns stripped,
+``run_nmfp.py:221-254``), CW frequency grid ``k/Tspan``, and output
+``{outdir}/{savefile}.npy`` of shape (nsamples, nfreqs).
+
+Additions over the reference: ``--outdir`` (the reference hard-codes
+``res/``), ``--device``, ``--seed``, checkpoint/resume of completed
+draw batches (``--resume``; per-batch shards are written to
+``{outdir}/.{savefile}.batches/``), and multi-GPU draw sharding via
+torchrun.
+"""
+
+import argparse
+import json
+import logging
+import os
+import time
+
+import numpy as np
+import torch
+
+from fastfp_amd.data import get_tspan, load_pulsars
+from fastfp_amd.engine import FpEngine
+from fastfp_amd.model import get_mats_nmfp, initialize_pta
+from fastfp_amd.nmfp import NMFp
+from fastfp_amd.parallel import (
+    all_gather_concat,
+    cleanup,
+    init_distributed,
+    shard_slice,
+)
+
+
+def setup_fp_model(psrs, noise, Tspan=None, add_ecorr=False, nrncomps=30,
+                   add_curn=False, ngwbcomps=5, pta=None):
+    """Build the NMFp object and its containers — signature parity with
+    the reference's ``setup_fp_model`` (``run_nmfp.py:73-171``).  Here
+    the containers already live on the PTAModel; reuse them when a
+    ``pta`` is passed, otherwise build a fresh model."""
+    if pta is None:
+        pta = initialize_pta(
+            psrs,
+            noise,
+            inc_cp=add_curn,
+            rn_comps=nrncomps,
+            gwb_comps=ngwbcomps,
+            inc_ecorr=add_ecorr,
+        )
+    return NMFp(psrs, pta.rn_containers)
+
+
+def map_params(pta, xs):
+    """2-D sample batch -> dict of per-parameter arrays (parity with
+    ``run_nmfp.py:174-186``)."""
+    xs = np.asarray(xs)
+    if xs.ndim > 1:
+        return {p: xs[ct, :] for ct, p in enumerate(pta.params)}
+    return pta.map_params(xs)
+
+
+def main(
+    psrfile,
+    noisefile,
+    chainfile,
+    savefile,
+    inc_ecorr=False,
+    inc_cp=False,
+    nrncomps=30,
+    ngwbcomps=30,
+    ncwfreqs=100,
+    nsamples=1000,
+    batch_size=100,
+    outdir="res",
+    device=None,
+    seed=0,
+    resume=False,
+):
+    logging.basicConfig(format="%(levelname)s: %(message)s", level=logging.INFO)
+    logger = logging.getLogger(__name__)
+
+    rank, world, dev = init_distributed(
+        device=torch.device(device) if device else None
+    )
+    logger.info(f"fastfp_amd backend device {dev} (rank {rank}/{world})")
+    logger.info(f"number of CW frequencies: {ncwfreqs}")
+    logger.info(f"number of samples: {nsamples}")
+    logger.info(f"batch_size: {batch_size}")
+
+    psrs = load_pulsars(psrfile)
+    with open(noisefile, "r") as f:
+        noise = json.load(f)
+
+    chain = np.loadtxt(chainfile)
+    if chain.ndim == 1:
+        chain = chain[None, :]
+    burn = int(0.25 * chain.shape[0])
+
+    noise["gw_gamma"] = 13 / 3
+    noise["gw_log10_A"] = float(np.log10(2e-15))
+
+    Tspan = get_tspan(psrs)
+    pta = initialize_pta(
+        psrs,
+        noise,
+        inc_cp=inc_cp,
+        rn_comps=nrncomps,
+        gwb_comps=ngwbcomps,
+        inc_ecorr=inc_ecorr,
+    )
+    nmfp = setup_fp_model(psrs, noise, pta=pta)
+
+    t0 = time.perf_counter()
+    TNTs, Nvecs, Ts = get_mats_nmfp(pta, noise)
+    logger.info(f"Precompute matrix wall time: {time.perf_counter() - t0:.2f} s")
+
+    freqs = np.arange(1, ncwfreqs + 1) / Tspan
+
+    # draw nsamples rows from the chain (after burn-in, bookkeeping
+    # columns stripped)
+    rng = np.random.default_rng(seed)
+    idxs = rng.choice(
+        np.arange(burn, chain.shape[0]),
+        size=min(nsamples, chain.shape[0] - burn),
+        replace=nsamples > chain.shape[0] - burn,
+    )
+    if len(idxs) < nsamples:
+        idxs = rng.choice(np.arange(burn, chain.shape[0]), nsamples, replace=True)
+    rns_full = chain[idxs, :-4].T  # (nparams, nsamples)
+
+    # shard draws across ranks
+    my = shard_slice(nsamples, rank, world)
+    my_idx = np.arange(nsamples)[my]
+
+    batch_dir = os.path.join(outdir, f".{savefile}.batches")
+    if rank == 0:
+        os.makedirs(outdir, exist_ok=True)
+        os.makedirs(batch_dir, exist_ok=True)
+    if world > 1:
+        torch.distributed.barrier()
+
+    eng = FpEngine(psrs, Nvecs, Ts, device=dev)
+    eng.precompute(freqs)
+
+    t0 = time.perf_counter()
+    parts = []
+    for lo in range(0, len(my_idx), batch_size):
+        sel = my_idx[lo : lo + batch_size]
+        ck = os.path.join(batch_dir, f"r{rank}_b{lo}.npy")
+        if resume and os.path.exists(ck):
+            parts.append(np.load(ck))
+            continue
+        samples = map_params(pta, rns_full[:, sel])
+        vals = nmfp.sweep(freqs, samples, Nvecs, Ts, engine=eng)
+        np.save(ck, vals)
+        parts.append(vals)
+    local_vals = (
+        np.vstack(parts) if parts else np.zeros((0, ncwfreqs))
+    )
+    t_local = torch.as_tensor(local_vals, dtype=torch.float64, device=dev)
+    full = all_gather_concat(t_local, world, dim=0).cpu().numpy()
+    logger.info(
+        f"Noise marginalized Fp-statistic wall time: {time.perf_counter() - t0:.2f} s"
+    )
+
+    if rank == 0:
+        with open(os.path.join(outdir, f"{savefile}.npy"), "wb") as f:
+            np.save(f, full)
+    cleanup()
+    return
+
+
+def cli():
+    parser = argparse.ArgumentParser(description=__doc__)
+    parser.add_argument("psrfile", type=str, help="pulsars file (.pkl/.npz/dir)")
+    parser.add_argument("noisefile", type=str, help="noise dictionary json")
+    parser.add_argument("chainfile", type=str, help="MCMC chain text file")
+    parser.add_argument("savefile", type=str, help="output .npy name (no ext)")
+    parser.add_argument("--inc_ecorr", action="store_true", help="include ECORR")
+    parser.add_argument("--inc_cp", action="store_true", help="include CURN process")
+    parser.add_argument("--nrncomps", type=int, default=30)
+    parser.add_argument("--ngwbcomps", type=int, default=30)
+    parser.add_argument("--ncwfreqs", type=int, default=100)
+    parser.add_argument("--nsamples", type=int, default=1000)
+    parser.add_argument("--batch_size", type=int, default=100)
+    parser.add_argument("--outdir", type=str, default="res")
+    parser.add_argument("--device", type=str, default=None)
+    parser.add_argument("--seed", type=int, default=0)
+    parser.add_argument("--resume", action="store_true",
+                        help="skip draw batches already on disk")
+    main(**vars(parser.parse_args()))
+
+
+if __name__ == "__main__":
+    cli()

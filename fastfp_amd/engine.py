@@ -180,13 +180,20 @@ class FpEngine:
                 if phiinvs is not None:
                     pinv = _t64(phiinvs[i], self.device)
                     pinv = pinv[None, :] if pinv.dim() == 1 else pinv[lo:hi]
-                    sigma = blk.TNT[None, :, :] + torch.diag_embed(pinv)
+                    sigma = None
                 else:
                     sg = _t64(sigmas[i], self.device)
                     sigma = sg[None, :, :] if sg.dim() == 2 else sg[lo:hi]
+                    # get_mats_fp contract: sigma = TNT + diag(phi^-1)
+                    pinv = (
+                        torch.diagonal(sigma, dim1=-2, dim2=-1)
+                        - torch.diagonal(blk.TNT)[None, :]
+                    )
                 if self._use_hip:
-                    self._accum_hip(blk, sigma, fp[lo:hi])
+                    self._accum_hip(blk, pinv.contiguous(), fp[lo:hi])
                 else:
+                    if sigma is None:
+                        sigma = blk.TNT[None, :, :] + torch.diag_embed(pinv)
                     self._accum_eager(blk, sigma, fp[lo:hi])
 
         return fp[0] if not batched else fp
@@ -214,9 +221,9 @@ class FpEngine:
         det = M11 * M22 - M12 * M12
         fp_out += 0.5 * (N1 * N1 * M22 - 2.0 * N1 * N2 * M12 + N2 * N2 * M11) / det
 
-    def _accum_hip(self, blk: PulsarBlock, sigma, fp_out):
+    def _accum_hip(self, blk: PulsarBlock, phiinv, fp_out):
         from fastfp_amd import ops
 
         ops.chol_trsm_fp_accum(
-            sigma, blk.RHS, blk.sNs, blk.sNr, fp_out
+            blk.TNT, phiinv, blk.RHS, blk.sNs, blk.sNr, fp_out
         )

@@ -48,50 +48,70 @@ def require_hip() -> None:
         )
 
 
+MAX_MP = 128  # kernel LDS-resident limit on the padded basis size
+
+
+def pad16(m: int) -> int:
+    return max(16, (m + 15) // 16 * 16)
+
+
+def check_m(m: int) -> int:
+    mp = pad16(m)
+    if mp > MAX_MP:
+        raise NotImplementedError(
+            f"GPU kernels currently support basis size m <= {MAX_MP} per "
+            f"pulsar (got m={m}); reduce the ECORR epoch count or use the "
+            "CPU engine for this model"
+        )
+    return mp
+
+
 # ----------------------------------------------------------------------
 # op wrappers (thin; shapes documented in the kernels)
 # ----------------------------------------------------------------------
 def freq_precompute(toas, Nvec, r, T, TNr, freqs, freq_chunk: int = 8192):
     """GPU frequency precompute for one pulsar.
 
-    Returns (RHS (m, 2F+1), sNs (3, F), sNr (2, F)) — the fused
-    sincos signal-basis kernel computes the diagonal-weighted dots and
-    the N^-1-scaled sin/cos panel; the MFMA DGEMM computes
-    B = T^T (N^-1 S).
+    Returns (RHS (mp, 2F+1), sNs (3, F), sNr (2, F)).  The fused
+    sigdots kernel computes the diagonal-weighted sin/cos dots; the
+    fused signal-basis MFMA DGEMM (sbgemm) computes B = T^T (N^-1 S)
+    with the S panel generated in LDS — NS is never materialized.
+    RHS rows are zero-padded to mp (multiple of 16).
     """
     ext = _try_load()
-    F = freqs.shape[0]
+    F = int(freqs.shape[0])
     ntoa, m = T.shape
+    mp = check_m(m)
     device = T.device
-    RHS = torch.empty((m, 2 * F + 1), dtype=torch.float64, device=device)
-    sNs = torch.empty((3, F), dtype=torch.float64, device=device)
-    sNr = torch.empty((2, F), dtype=torch.float64, device=device)
-    Nr = r / Nvec
-    Ninv = 1.0 / Nvec
-    for lo in range(0, F, freq_chunk):
-        hi = min(lo + freq_chunk, F)
-        fchunk = freqs[lo:hi].contiguous()
-        # NS panel (2Fc, ntoa) row-major + dots, one fused kernel
-        NS, sNs_c, sNr_c = ext.sigbasis(toas, Ninv, Nr, fchunk)
-        sNs[:, lo:hi] = sNs_c
-        sNr[:, lo:hi] = sNr_c
-        # B = NS @ T -> (2Fc, m); store transposed into RHS columns
-        Bc = ext.dgemm_nn(NS, T)  # (2Fc, m)
-        RHS[:, 2 * lo : 2 * hi].copy_(Bc.transpose(0, 1))
-    RHS[:, -1] = TNr
+    Ninv = (1.0 / Nvec).contiguous()
+    Nr = (r / Nvec).contiguous()
+    freqs = freqs.contiguous()
+
+    sNs, sNr = ext.sigdots(toas.contiguous(), Ninv, Nr, freqs)
+
+    RHS = torch.zeros((mp, 2 * F + 1), dtype=torch.float64, device=device)
+    F2 = 2 * F
+    ctiles = (F2 + 63) // 64
+    # split K to fill the 256-CU chip when the frequency grid is small;
+    # partial planes + a deterministic torch reduction keep fp64
+    # bitwise-reproducible (no atomics)
+    ksplit = max(1, min(8, (512 + ctiles - 1) // ctiles, (ntoa + 255) // 256))
+    if ksplit == 1:
+        ext.sbgemm(T, toas, Ninv, freqs, RHS, 0, 2 * F + 1, mp, 1)
+    else:
+        part = torch.empty((ksplit, mp, F2), dtype=torch.float64, device=device)
+        ext.sbgemm(T, toas, Ninv, freqs, part, mp * F2, F2, mp, ksplit)
+        RHS[:, :F2] = part.sum(dim=0)
+    RHS[:m, -1] = TNr
     return RHS, sNs, sNr
 
 
-def chol_trsm_fp_accum(sigma, RHS, sNs, sNr, fp_out):
-    """Batched Cholesky of sigma (D, m, m) in place, then fused
-    triangular solve of RHS (m, 2F+1) + 2x2 Fp reduction, accumulating
-    into fp_out (D, F)."""
+def chol_trsm_fp_accum(TNT, phiinv, RHS, sNs, sNr, fp_out):
+    """Batched Cholesky of Sigma = TNT + diag(phiinv) (assembled in
+    LDS), then the fused triangular solve of RHS (mp, 2F+1) + 2x2 Fp
+    reduction, accumulating into fp_out (D, F)."""
     ext = _try_load()
-    ext.chol_batch(sigma)
-    ext.trsm_fp_accum(sigma, RHS, sNs, sNr, fp_out)
-
-
-def dgemm_tn(A, B):
-    """C = A^T @ B for fp64 (K, M) x (K, N) -> (M, N)."""
-    ext = _try_load()
-    return ext.dgemm_tn(A, B)
+    m = TNT.shape[0]
+    mp = check_m(m)
+    L, invd = ext.chol_batch(TNT.contiguous(), phiinv.contiguous(), mp)
+    ext.trsm_fp_accum(L, invd, RHS, sNs, sNr, fp_out)

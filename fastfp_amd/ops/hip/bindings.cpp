@@ -1,0 +1,122 @@
+// PyTorch bindings for the fastfp_amd fp64 HIP kernels (MI355X/gfx950).
+//
+// Thin shape/contiguity-checked wrappers; all launches go onto the
+// current torch HIP stream so the ops compose with torch eager code.
+
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+#include <hip/hip_runtime.h>
+
+extern "C" {
+void launch_sigdots(const double*, const double*, const double*,
+                    const double*, int, int, double*, double*, hipStream_t);
+void launch_sbgemm(const double*, const double*, const double*,
+                   const double*, int, int, int, int, double*, long, long,
+                   int, hipStream_t);
+void launch_chol_batch(const double*, const double*, int, int, int, double*,
+                       double*, hipStream_t);
+void launch_trsm_fp(const double*, const double*, const double*,
+                    const double*, const double*, int, int, int, double*,
+                    hipStream_t);
+}
+
+namespace {
+
+void check_f64(const torch::Tensor& t, const char* name) {
+  TORCH_CHECK(t.is_cuda(), name, " must be on the GPU");
+  TORCH_CHECK(t.scalar_type() == torch::kFloat64, name, " must be fp64");
+  TORCH_CHECK(t.is_contiguous(), name, " must be contiguous");
+}
+
+hipStream_t stream() {
+  return at::hip::getCurrentHIPStream().stream();
+}
+
+}  // namespace
+
+// sigdots: (toas, ninv, nr (ntoa,), freqs (F,)) -> sNs (3,F), sNr (2,F)
+std::vector<torch::Tensor> sigdots(torch::Tensor toas, torch::Tensor ninv,
+                                   torch::Tensor nr, torch::Tensor freqs) {
+  check_f64(toas, "toas");
+  check_f64(ninv, "ninv");
+  check_f64(nr, "nr");
+  check_f64(freqs, "freqs");
+  const int ntoa = toas.size(0);
+  const int F = freqs.size(0);
+  auto opts = toas.options();
+  auto sNs = torch::empty({3, F}, opts);
+  auto sNr = torch::empty({2, F}, opts);
+  launch_sigdots(toas.data_ptr<double>(), ninv.data_ptr<double>(),
+                 nr.data_ptr<double>(), freqs.data_ptr<double>(), ntoa, F,
+                 sNs.data_ptr<double>(), sNr.data_ptr<double>(), stream());
+  return {sNs, sNr};
+}
+
+// sbgemm into `out` (plane-strided), out[j, col0 + c] over F2 columns.
+void sbgemm(torch::Tensor T, torch::Tensor toas, torch::Tensor ninv,
+            torch::Tensor freqs, torch::Tensor out, int64_t plane_stride,
+            int64_t ldo, int64_t mp, int64_t ksplit) {
+  check_f64(T, "T");
+  check_f64(toas, "toas");
+  check_f64(ninv, "ninv");
+  check_f64(freqs, "freqs");
+  check_f64(out, "out");
+  const int ntoa = T.size(0);
+  const int m = T.size(1);
+  const int F2 = 2 * freqs.size(0);
+  TORCH_CHECK(mp % 16 == 0 && mp <= 128 && mp >= m, "bad mp");
+  launch_sbgemm(T.data_ptr<double>(), toas.data_ptr<double>(),
+                ninv.data_ptr<double>(), freqs.data_ptr<double>(), ntoa, m,
+                (int)mp, F2, out.data_ptr<double>(), plane_stride, ldo,
+                (int)ksplit, stream());
+}
+
+// chol_batch: TNT (m,m), phiinv (D,m) -> L (D,mp,mp), invd (D,mp/16,16,16)
+std::vector<torch::Tensor> chol_batch(torch::Tensor TNT, torch::Tensor phiinv,
+                                      int64_t mp) {
+  check_f64(TNT, "TNT");
+  check_f64(phiinv, "phiinv");
+  const int m = TNT.size(0);
+  const int D = phiinv.size(0);
+  TORCH_CHECK(phiinv.size(1) == m, "phiinv width != m");
+  TORCH_CHECK(mp % 16 == 0 && mp <= 128 && mp >= m,
+              "chol_batch requires m <= 128 (basis size per pulsar); "
+              "got m=", m);
+  auto opts = TNT.options();
+  auto L = torch::empty({D, mp, mp}, opts);
+  auto invd = torch::empty({D, mp / 16, 16, 16}, opts);
+  launch_chol_batch(TNT.data_ptr<double>(), phiinv.data_ptr<double>(), m,
+                    (int)mp, D, L.data_ptr<double>(),
+                    invd.data_ptr<double>(), stream());
+  return {L, invd};
+}
+
+// trsm_fp_accum: L (D,mp,mp), invd, RHS (mp, 2F+1), sNs (3,F), sNr (2,F),
+// fp (D,F) accumulated in place.
+void trsm_fp_accum(torch::Tensor L, torch::Tensor invd, torch::Tensor RHS,
+                   torch::Tensor sNs, torch::Tensor sNr, torch::Tensor fp) {
+  check_f64(L, "L");
+  check_f64(invd, "invd");
+  check_f64(RHS, "RHS");
+  check_f64(sNs, "sNs");
+  check_f64(sNr, "sNr");
+  check_f64(fp, "fp");
+  const int D = L.size(0);
+  const int mp = L.size(1);
+  const int F = sNs.size(1);
+  TORCH_CHECK(RHS.size(0) == mp && RHS.size(1) == 2 * F + 1, "RHS shape");
+  TORCH_CHECK(fp.size(0) == D && fp.size(1) == F, "fp shape");
+  TORCH_CHECK(mp % 16 == 0 && mp <= 128, "mp must be <=128, multiple of 16");
+  launch_trsm_fp(L.data_ptr<double>(), invd.data_ptr<double>(),
+                 RHS.data_ptr<double>(), sNs.data_ptr<double>(),
+                 sNr.data_ptr<double>(), mp, F, D, fp.data_ptr<double>(),
+                 stream());
+}
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("sigdots", &sigdots, "fused sincos signal-basis dots");
+  m.def("sbgemm", &sbgemm, "fused signal-basis MFMA fp64 DGEMM");
+  m.def("chol_batch", &chol_batch, "batched LDS-resident fp64 Cholesky");
+  m.def("trsm_fp_accum", &trsm_fp_accum,
+        "batched TRSM + fused 2x2 Fp reduction");
+}

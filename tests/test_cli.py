@@ -1,0 +1,79 @@
+"""CLI round-trip tests: input loading, output formats (JSON / .npy),
+resume behavior — the reference's exact output contracts
+(/root/reference/examples/run_fp.py:69-71, run_nmfp.py:270-276)."""
+
+import json
+import os
+
+import numpy as np
+import pytest
+
+from fastfp_amd.cli import run_fp, run_nmfp
+from fastfp_amd.data import make_synthetic_pta, save_pulsars
+
+
+@pytest.fixture(scope="module")
+def inputs(tmp_path_factory):
+    tmp = tmp_path_factory.mktemp("cli")
+    psrs = make_synthetic_pta(npsr=2, ntoa=60, ntm=3, seed=0)
+    psrfile = str(tmp / "psrs.npz")
+    save_pulsars(psrs, psrfile)
+    noise = {}
+    for p in psrs:
+        noise[f"{p.name}_red_noise_gamma"] = 4.0
+        noise[f"{p.name}_red_noise_log10_A"] = -14.5
+    noisefile = str(tmp / "noise.json")
+    with open(noisefile, "w") as f:
+        json.dump(noise, f)
+    # fake chain: params sorted (2 per psr + 2 gw) + 4 bookkeeping cols
+    nparams = 2 * len(psrs) + 2
+    rng = np.random.default_rng(1)
+    chain = np.zeros((40, nparams + 4))
+    for i in range(nparams):
+        chain[:, i] = rng.uniform(2.0, 6.0, 40) if i % 2 == 0 else rng.uniform(-16, -14, 40)
+    # columns alternate by sorted name: gamma cols then log10_A cols per psr;
+    # order doesn't matter for smoke purposes as long as values are sane
+    chainfile = str(tmp / "chain.txt")
+    np.savetxt(chainfile, chain)
+    return tmp, psrfile, noisefile, chainfile
+
+
+def test_run_fp_json_output(inputs):
+    tmp, psrfile, noisefile, _ = inputs
+    out = str(tmp / "fpout")
+    run_fp.main(psrfile, noisefile, out, nfreqs=5, rn_comps=3, gwb_comps=3,
+                device="cpu")
+    with open(out + ".json") as f:
+        res = json.load(f)
+    assert len(res) == 5
+    freqs = np.array(sorted(float(k) for k in res))
+    np.testing.assert_allclose(freqs, np.linspace(2e-9, 3e-7, 5))
+    assert all(np.isfinite(v) for v in res.values())
+
+
+def test_run_nmfp_npy_output(inputs):
+    tmp, psrfile, noisefile, chainfile = inputs
+    outdir = str(tmp / "res")
+    run_nmfp.main(
+        psrfile, noisefile, chainfile, "nm",
+        inc_cp=True, nrncomps=3, ngwbcomps=3, ncwfreqs=4,
+        nsamples=6, batch_size=4, outdir=outdir, device="cpu",
+    )
+    vals = np.load(os.path.join(outdir, "nm.npy"))
+    assert vals.shape == (6, 4)
+    assert np.isfinite(vals).all()
+
+
+def test_run_nmfp_resume(inputs):
+    tmp, psrfile, noisefile, chainfile = inputs
+    outdir = str(tmp / "res2")
+    kwargs = dict(
+        inc_cp=False, nrncomps=3, ncwfreqs=3, nsamples=4, batch_size=2,
+        outdir=outdir, device="cpu", seed=3,
+    )
+    run_nmfp.main(psrfile, noisefile, chainfile, "nm", **kwargs)
+    a = np.load(os.path.join(outdir, "nm.npy"))
+    # resume run must reproduce identical output from the batch shards
+    run_nmfp.main(psrfile, noisefile, chainfile, "nm", resume=True, **kwargs)
+    b = np.load(os.path.join(outdir, "nm.npy"))
+    np.testing.assert_array_equal(a, b)

@@ -1,0 +1,202 @@
+"""GPU kernel numerics: every HIP kernel vs a plain PyTorch fp64
+reference on-device, plus end-to-end HIP-vs-eager and
+bitwise-determinism checks."""
+
+import math
+
+import numpy as np
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+DEV = "cuda:0"
+
+
+def _ext():
+    from fastfp_amd import ops
+
+    ops.require_hip()
+    from fastfp_amd.ops import _fastfp_hip
+
+    return _fastfp_hip
+
+
+def _rand_problem(seed=0, ntoa=333, m=37, F=29, D=5):
+    rng = np.random.default_rng(seed)
+    toas = torch.as_tensor(
+        np.sort(rng.uniform(0, 4e8, ntoa)), dtype=torch.float64, device=DEV
+    )
+    nvec = torch.as_tensor(
+        rng.uniform(0.5, 2.0, ntoa) * 1e-12, dtype=torch.float64, device=DEV
+    )
+    r = torch.as_tensor(
+        rng.normal(0, 1e-6, ntoa), dtype=torch.float64, device=DEV
+    )
+    T = torch.as_tensor(
+        rng.normal(size=(ntoa, m)), dtype=torch.float64, device=DEV
+    )
+    freqs = torch.as_tensor(
+        np.linspace(3e-9, 5e-8, F), dtype=torch.float64, device=DEV
+    )
+    phiinv = torch.as_tensor(
+        rng.uniform(0.5, 2.0, (D, m)) * 1e10, dtype=torch.float64, device=DEV
+    )
+    return toas, nvec, r, T, freqs, phiinv
+
+
+def test_sigdots_vs_torch():
+    ext = _ext()
+    toas, nvec, r, T, freqs, _ = _rand_problem()
+    ninv = 1.0 / nvec
+    nr = r / nvec
+    sNs, sNr = ext.sigdots(toas, ninv.contiguous(), nr.contiguous(), freqs)
+    arg = 2 * math.pi * freqs[:, None] * toas[None, :]
+    S, C = torch.sin(arg), torch.cos(arg)
+    torch.testing.assert_close(sNs[0], (S * S * ninv).sum(1), rtol=1e-12, atol=1e-6)
+    torch.testing.assert_close(sNs[1], (C * C * ninv).sum(1), rtol=1e-12, atol=1e-6)
+    torch.testing.assert_close(sNs[2], (S * C * ninv).sum(1), rtol=1e-10, atol=1e-4)
+    torch.testing.assert_close(sNr[0], S @ nr, rtol=1e-10, atol=1e-8)
+    torch.testing.assert_close(sNr[1], C @ nr, rtol=1e-10, atol=1e-8)
+
+
+@pytest.mark.parametrize("F,ntoa", [(29, 333), (70, 150), (200, 2000)])
+def test_sbgemm_vs_torch(F, ntoa):
+    from fastfp_amd import ops
+
+    toas, nvec, r, T, freqs, _ = _rand_problem(F=F, ntoa=ntoa)
+    ninv = 1.0 / nvec
+    TNr = (T / nvec[:, None]).T @ r
+    RHS, sNs, sNr = ops.freq_precompute(toas, nvec, r, T, TNr, freqs)
+    m = T.shape[1]
+    mp = RHS.shape[0]
+    assert mp % 16 == 0
+    arg = 2 * math.pi * freqs[:, None] * toas[None, :]
+    S = torch.sin(arg) * ninv[None, :]
+    C = torch.cos(arg) * ninv[None, :]
+    want_s = S @ T  # (F, m)
+    want_c = C @ T
+    torch.testing.assert_close(RHS[:m, 0:-1:2], want_s.T, rtol=1e-11, atol=1e-9)
+    torch.testing.assert_close(RHS[:m, 1:-1:2], want_c.T, rtol=1e-11, atol=1e-9)
+    assert (RHS[m:, :] == 0).all()
+    torch.testing.assert_close(RHS[:m, -1], TNr, rtol=1e-12, atol=0.0)
+
+
+def test_chol_batch_vs_torch():
+    ext = _ext()
+    toas, nvec, r, T, freqs, phiinv = _rand_problem()
+    m = T.shape[1]
+    mp = 48  # padded
+    TN = T / nvec[:, None]
+    TNT = (T.T @ TN).contiguous()
+    L, invd = ext.chol_batch(TNT, phiinv, mp)
+    D = phiinv.shape[0]
+    sigma = TNT[None] + torch.diag_embed(phiinv)
+    want = torch.linalg.cholesky(sigma)
+    got = torch.tril(L[:, :m, :m])
+    torch.testing.assert_close(got, want, rtol=1e-9, atol=1e-9)
+    # pad block identity
+    assert (torch.tril(L[:, m:, m:]) == torch.eye(mp - m, device=DEV)).all()
+    # inverted diagonal blocks
+    nb = mp // 16
+    for kb in range(nb):
+        blk = torch.tril(L[:, kb * 16 : kb * 16 + 16, kb * 16 : kb * 16 + 16])
+        inv = torch.linalg.solve_triangular(
+            blk, torch.eye(16, dtype=torch.float64, device=DEV)[None].expand(D, 16, 16),
+            upper=False,
+        )
+        torch.testing.assert_close(
+            torch.tril(invd[:, kb]), torch.tril(inv), rtol=1e-8, atol=1e-8
+        )
+
+
+def test_trsm_fp_accum_vs_eager():
+    from fastfp_amd import ops
+
+    toas, nvec, r, T, freqs, phiinv = _rand_problem(F=100, D=7)
+    m = T.shape[1]
+    F = freqs.shape[0]
+    D = phiinv.shape[0]
+    TN = T / nvec[:, None]
+    TNT = (T.T @ TN).contiguous()
+    TNr = TN.T @ r
+    RHS, sNs, sNr = ops.freq_precompute(toas, nvec, r, T, TNr, freqs)
+
+    fp = torch.zeros((D, F), dtype=torch.float64, device=DEV)
+    ops.chol_trsm_fp_accum(TNT, phiinv, RHS, sNs, sNr, fp)
+
+    # eager reference (same restructured math, torch linalg)
+    sigma = TNT[None] + torch.diag_embed(phiinv)
+    Lr = torch.linalg.cholesky(sigma)
+    RHSe = torch.cat([RHS[:m, :-1], TNr[:, None]], dim=1)
+    W = torch.linalg.solve_triangular(
+        Lr, RHSe[None].expand(D, -1, -1), upper=False
+    )
+    wu = W[:, :, -1]
+    Ws, Wc = W[:, :, 0:-1:2], W[:, :, 1:-1:2]
+    M11 = sNs[0][None] - (Ws * Ws).sum(1)
+    M22 = sNs[1][None] - (Wc * Wc).sum(1)
+    M12 = sNs[2][None] - (Ws * Wc).sum(1)
+    N1 = sNr[0][None] - torch.einsum("dmf,dm->df", Ws, wu)
+    N2 = sNr[1][None] - torch.einsum("dmf,dm->df", Wc, wu)
+    det = M11 * M22 - M12 * M12
+    want = 0.5 * (N1 * N1 * M22 - 2 * N1 * N2 * M12 + N2 * N2 * M11) / det
+    torch.testing.assert_close(fp, want, rtol=1e-7, atol=1e-9)
+
+    # accumulation: second call doubles
+    fp2 = fp.clone()
+    ops.chol_trsm_fp_accum(TNT, phiinv, RHS, sNs, sNr, fp2)
+    torch.testing.assert_close(fp2, 2 * fp, rtol=1e-12, atol=0.0)
+
+
+def test_engine_gpu_matches_cpu_eager():
+    from fastfp_amd import FpEngine, get_mats_nmfp, initialize_pta, make_synthetic_pta
+    from fastfp_amd.nmfp import NMFp
+
+    psrs = make_synthetic_pta(npsr=3, ntoa=500, ntm=5, seed=3)
+    noise = {"gw_gamma": 13.0 / 3.0, "gw_log10_A": float(np.log10(2e-15))}
+    for p in psrs:
+        noise[f"{p.name}_red_noise_gamma"] = 4.0
+        noise[f"{p.name}_red_noise_log10_A"] = -14.5
+    pta = initialize_pta(psrs, noise, inc_cp=True, rn_comps=6, gwb_comps=6)
+    TNTs, Nvecs, Ts = get_mats_nmfp(pta, noise)
+    D = 4
+    rng = np.random.default_rng(7)
+    samples = {}
+    for name in pta.params:
+        samples[name] = (
+            rng.uniform(2, 6, D) if name.endswith("gamma") else rng.uniform(-16, -14, D)
+        )
+    nm = NMFp(psrs, pta.rn_containers)
+    freqs = np.linspace(4e-9, 6e-8, 33)
+    cpu = nm.sweep(freqs, samples, Nvecs, Ts, device="cpu")
+    for c in pta.rn_containers:
+        c.to(DEV)
+    gpu = nm.sweep(freqs, samples, Nvecs, Ts, device=DEV)
+    np.testing.assert_allclose(gpu, cpu, rtol=1e-8)
+
+
+def test_gpu_determinism_bitwise():
+    from fastfp_amd import ops
+
+    toas, nvec, r, T, freqs, phiinv = _rand_problem(F=65, D=6)
+    TN = T / nvec[:, None]
+    TNT = (T.T @ TN).contiguous()
+    TNr = TN.T @ r
+    outs = []
+    for _ in range(2):
+        RHS, sNs, sNr = ops.freq_precompute(toas, nvec, r, T, TNr, freqs)
+        fp = torch.zeros((phiinv.shape[0], freqs.shape[0]), dtype=torch.float64, device=DEV)
+        ops.chol_trsm_fp_accum(TNT, phiinv, RHS, sNs, sNr, fp)
+        outs.append(fp.cpu().numpy())
+    np.testing.assert_array_equal(outs[0], outs[1])
+
+
+def test_native_extension_is_loaded():
+    """Guard against silent eager fallback on GPU boxes."""
+    from fastfp_amd import ops
+
+    assert ops.hip_available(), "HIP extension must be importable on a GPU box"
+    from fastfp_amd.ops import _fastfp_hip
+
+    assert "fastfp_amd" in _fastfp_hip.__file__
