@@ -30,7 +30,7 @@
 //   wave = 64 lanes (CDNA);  MFMA v_mfma_f64_16x16x4f64:
 //     A[i][k]: lane l holds A[i = l&15][k = l>>4]
 //     B[k][j]: lane l holds B[k = l>>4][j = l&15]
-//     D/C    : lane l holds rows (l>>4)*4 + v (v=0..3), col l&15
+//     D/C    : lane l holds rows 4*v + (l>>4) (v=0..3), col l&15 (probed: tools/mfma_probe.hip)
 //   mp = m padded to a multiple of 16, mp <= 128.  Pad rows of Sigma
 //   are identity (L pad = I) and pad rows of RHS are zero, so padding
 //   never changes results.
@@ -181,7 +181,7 @@ extern "C" __global__ __launch_bounds__(256) void sbgemm_kernel(
     if (rt >= nrt) break;
 #pragma unroll
     for (int v = 0; v < 4; ++v) {
-      const int row = rt * 16 + (lane >> 4) * 4 + v;
+      const int row = rt * 16 + 4 * v + (lane >> 4);
       const int col = c0 + jw + (lane & 15);
       if (col < F2) outp[(long)row * ldo + col] = acc[rt][v];
     }
@@ -277,7 +277,7 @@ extern "C" __global__ __launch_bounds__(256) void chol_batch_kernel(
       // in-wave: all reads above complete before these writes
 #pragma unroll
       for (int v = 0; v < 4; ++v)
-        A_(r0 + (lane >> 4) * 4 + v, k0 + (lane & 15)) = pacc[v];
+        A_(r0 + 4 * v + (lane >> 4), k0 + (lane & 15)) = pacc[v];
     }
     __syncthreads();
 
@@ -293,7 +293,7 @@ extern "C" __global__ __launch_bounds__(256) void chol_batch_kernel(
       f64x4 uacc;
 #pragma unroll
       for (int v = 0; v < 4; ++v)
-        uacc[v] = A_(i0 + (lane >> 4) * 4 + v, j0 + (lane & 15));
+        uacc[v] = A_(i0 + 4 * v + (lane >> 4), j0 + (lane & 15));
 #pragma unroll
       for (int kk = 0; kk < 4; ++kk) {
         const double a = -A_(i0 + (lane & 15), k0 + kk * 4 + (lane >> 4));
@@ -302,7 +302,7 @@ extern "C" __global__ __launch_bounds__(256) void chol_batch_kernel(
       }
 #pragma unroll
       for (int v = 0; v < 4; ++v)
-        A_(i0 + (lane >> 4) * 4 + v, j0 + (lane & 15)) = uacc[v];
+        A_(i0 + 4 * v + (lane >> 4), j0 + (lane & 15)) = uacc[v];
     }
     __syncthreads();
   }
@@ -380,7 +380,7 @@ extern "C" __global__ __launch_bounds__(512) void trsm_fp_kernel(
     f64x4 acc;
 #pragma unroll
     for (int v = 0; v < 4; ++v)
-      acc[v] = -W_(r0 + (lane >> 4) * 4 + v, jw + (lane & 15));
+      acc[v] = -W_(r0 + 4 * v + (lane >> 4), jw + (lane & 15));
     for (int cb = 0; cb < rb; ++cb) {
 #pragma unroll
       for (int kk = 0; kk < 4; ++kk) {
@@ -392,7 +392,7 @@ extern "C" __global__ __launch_bounds__(512) void trsm_fp_kernel(
     __syncthreads();  // everyone done READING W[rb] (acc init) before overwrite
 #pragma unroll
     for (int v = 0; v < 4; ++v)
-      W_(r0 + (lane >> 4) * 4 + v, jw + (lane & 15)) = acc[v];
+      W_(r0 + 4 * v + (lane >> 4), jw + (lane & 15)) = acc[v];
     __syncthreads();
     // W[rb] <- inv(L_rb,rb) * (RHS - sum) = Iv * (-Wtmp)
     f64x4 sol = {0, 0, 0, 0};
@@ -405,7 +405,7 @@ extern "C" __global__ __launch_bounds__(512) void trsm_fp_kernel(
     __syncthreads();
 #pragma unroll
     for (int v = 0; v < 4; ++v)
-      W_(r0 + (lane >> 4) * 4 + v, jw + (lane & 15)) = sol[v];
+      W_(r0 + 4 * v + (lane >> 4), jw + (lane & 15)) = sol[v];
     __syncthreads();
   }
 
