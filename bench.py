@@ -33,6 +33,7 @@ import torch
 from fastfp_amd.data import make_synthetic_pta
 from fastfp_amd.engine import FpEngine
 from fastfp_amd.model import get_mats_nmfp, initialize_pta
+from fastfp_amd.noise import batch_phiinv, check_batch_homogeneous
 from fastfp_amd.parallel import all_gather_concat, cleanup, init_distributed
 
 
@@ -72,11 +73,13 @@ def build_problem(args, device, rank):
         else:
             v = rng.uniform(-16.0, -13.5, D)
         pool[name] = torch.as_tensor(v, dtype=torch.float64, device=device)
+    # syncing check, done once here so the captured step never syncs
+    pta._phi_homog = check_batch_homogeneous(pta.rn_containers)
     return pta, eng, pool
 
 
 def run_step(pta, eng, pool, args, fp_accum):
-    phiinvs = [cont.get_phiinv(pool) for cont in pta.rn_containers]
+    phiinvs = batch_phiinv(pta.rn_containers, pool, homogeneous=pta._phi_homog)
     fp_accum.zero_()
     eng.sweep(phiinvs=phiinvs, draw_chunk=args.draw_chunk, accumulate_to=fp_accum)
     return fp_accum
@@ -94,9 +97,11 @@ def main():
     ap.add_argument("--gwb-comps", type=int, default=30)
     ap.add_argument("--freqs", type=int, default=1000)
     ap.add_argument("--draws-per-step", type=int, default=500)
-    ap.add_argument("--draw-chunk", type=int, default=64)
+    ap.add_argument("--draw-chunk", type=int, default=512)
     ap.add_argument("--freq-chunk", type=int, default=4096)
     ap.add_argument("--device", type=str, default=None)
+    ap.add_argument("--no-graph", action="store_true",
+                    help="disable hipGraph capture of the step")
     args = ap.parse_args()
 
     rank, world, device = init_distributed(
@@ -115,13 +120,28 @@ def main():
         if on_gpu:
             torch.cuda.synchronize()
 
+    # hipGraph-capture the whole step (67 pulsars x ~2 kernel launches
+    # + the batched phi assembly): launch gaps between the many small
+    # dispatches otherwise cost ~15% of the step
+    step = lambda: run_step(pta, eng, pool, args, fp_accum)  # noqa: E731
+    if on_gpu and not args.no_graph:
+        side = torch.cuda.Stream()
+        side.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(side):
+            step()
+        torch.cuda.current_stream().wait_stream(side)
+        graph = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(graph):
+            run_step(pta, eng, pool, args, fp_accum)
+        step = graph.replay
+
     for _ in range(args.warmup):
-        run_step(pta, eng, pool, args, fp_accum)
+        step()
 
     barrier_sync()
     t0 = time.perf_counter()
     for _ in range(args.steps):
-        run_step(pta, eng, pool, args, fp_accum)
+        step()
     # the spectrum all-gather is part of the job
     full = all_gather_concat(fp_accum, world, dim=0)
     barrier_sync()

@@ -314,3 +314,76 @@ class RNContainer:
         if self.gp_ecorr:
             n += self.ecorr_container.get_phi().shape[0]
         return n
+
+
+def check_batch_homogeneous(containers: list) -> bool:
+    """True when all containers share the Fourier grid and tm size and
+    have no ECORR (the batchable case).  Device-syncing comparison —
+    call ONCE outside any hipGraph capture or hot loop."""
+    c0 = containers[0]
+    return all(
+        (not c.gp_ecorr)
+        and c.inc_tm
+        and c.tm_weights.shape[0] == c0.tm_weights.shape[0]
+        and c.Ffreqs.shape[0] == c0.Ffreqs.shape[0]
+        and bool(torch.equal(c.Ffreqs, c0.Ffreqs))
+        and c.add_curn == c0.add_curn
+        for c in containers
+    )
+
+
+def batch_phiinv(containers: list, pars: dict, homogeneous: bool = None):
+    """Draw-vectorized phi^-1 for MANY pulsars in one fused op chain.
+
+    When every container shares the same Fourier grid, tm size, and has
+    no ECORR (the homogeneous case of the benchmark configs), the
+    per-pulsar power laws batch into a single (P, D, m) computation —
+    ~6 torch kernels instead of ~15 per pulsar per step.  Falls back to
+    the per-container path otherwise.  Returns a list of (D, m) views.
+
+    ``homogeneous``: pass the result of :func:`check_batch_homogeneous`
+    when calling from a hipGraph capture (the check itself syncs).
+    """
+    if homogeneous is None:
+        homogeneous = check_batch_homogeneous(containers)
+    if not homogeneous:
+        return [c.get_phiinv(pars) for c in containers]
+    c0 = containers[0]
+
+    dev = c0.Ffreqs.device
+    Ff = c0.Ffreqs
+    gam = torch.stack(
+        [_as_tensor(pars[c.rn_gam_name], dev).reshape(-1) for c in containers]
+    )  # (P, D)
+    lgA = torch.stack(
+        [_as_tensor(pars[c.rn_A_name], dev).reshape(-1) for c in containers]
+    )
+    P, D = gam.shape
+    funique = Ff[::2]
+    df = torch.diff(
+        torch.cat([torch.zeros(1, dtype=Ff.dtype, device=dev), funique])
+    )
+    dff = torch.repeat_interleave(df, 2)
+    phi_rn = (
+        Ff[None, None, :] ** (-gam[:, :, None])
+        * (10.0 ** lgA[:, :, None]) ** 2
+        / 12.0
+        / np.pi**2
+        * fyr ** (gam[:, :, None] - 3.0)
+        * dff[None, None, :]
+    )  # (P, D, nf)
+    if c0.add_curn:
+        curn = c0.curn_container.get_phi_curn(pars)  # (D, nc) or (nc,)
+        if curn.dim() == 1:
+            curn = curn[None, :].expand(D, -1)
+        nc = curn.shape[-1]
+        phi_rn[:, :, :nc] = phi_rn[:, :, :nc] + curn[None, :, :]
+    ntm = c0.tm_weights.shape[0]
+    out = torch.empty(
+        (P, D, ntm + Ff.shape[0]), dtype=torch.float64, device=dev
+    )
+    out[:, :, :ntm] = 1.0 / TM_PRIOR
+    out[:, :, ntm:] = 1.0 / phi_rn
+    return [out[i] for i in range(P)]
+
+

@@ -193,11 +193,14 @@ extern "C" __global__ __launch_bounds__(256) void sbgemm_kernel(
 // identity padding to mp) in LDS, factor L L^T = Sigma (right-looking,
 // NB=16 blocks, MFMA trailing updates), invert each 16x16 diagonal
 // block, write L (D, mp, mp) and invdiag (D, mp/16, 16, 16).
-// grid.x = D, block = 256 (4 waves)
+// grid.x = D, block = 512 (8 waves).  The 16x16 diagonal factor and its
+// inversion run REGISTER-resident in wave 0 with cross-lane __shfl
+// (width 16) -- the LDS read-modify-write chains of the naive version
+// were the kernel's dominant cost (profiles/r01_initial_stats.md).
 // ---------------------------------------------------------------------
 #define A_(i, j) Ash[(i) * (FASTFP_MAXMP + 1) + (j)]
 
-extern "C" __global__ __launch_bounds__(256) void chol_batch_kernel(
+extern "C" __global__ __launch_bounds__(512) void chol_batch_kernel(
     const double* __restrict__ TNT /*(m,m)*/,
     const double* __restrict__ phiinv /*(D,m)*/, int m, int mp, int D,
     double* __restrict__ L /*(D,mp,mp)*/,
@@ -213,7 +216,7 @@ extern "C" __global__ __launch_bounds__(256) void chol_batch_kernel(
   const int nb = mp >> 4;
 
   // assemble Sigma in LDS
-  for (int idx = tid; idx < mp * mp; idx += 256) {
+  for (int idx = tid; idx < mp * mp; idx += 512) {
     const int i = idx / mp, j = idx % mp;
     double v = (i < m && j < m) ? TNT[(long)i * m + j] : 0.0;
     if (i == j) v += (i < m) ? phiinv[(long)d * m + i] : 1.0;
@@ -224,48 +227,58 @@ extern "C" __global__ __launch_bounds__(256) void chol_batch_kernel(
   for (int kb = 0; kb < nb; ++kb) {
     const int k0 = kb * NB;
 
-    // 1) unblocked Cholesky of the 16x16 diagonal block (wave 0,
-    //    lanes 0..15 own rows; in-wave lockstep ordering)
-    if (wv == 0 && lane < 16) {
-      const int i = lane;
+    // 1) register-resident Cholesky of the 16x16 diagonal block.
+    //    Wave 0: lane segment s (16 lanes) redundantly owns rows i=l&15;
+    //    all cross-row traffic is __shfl (width 16), no LDS in the chain.
+    if (wv == 0) {
+      const int i = lane & 15;
+      double row[16];
+#pragma unroll
+      for (int c = 0; c < 16; ++c) row[c] = A_(k0 + i, k0 + c);
+#pragma unroll
       for (int t = 0; t < 16; ++t) {
-        const double att = A_(k0 + t, k0 + t);
-        const double dv = sqrt(att);
-        double lit = 0.0;
-        if (i > t) {
-          lit = A_(k0 + i, k0 + t) / dv;
-          A_(k0 + i, k0 + t) = lit;
+        const double dv = sqrt(__shfl(row[t], t, 16));
+        const double rdv = 1.0 / dv;  // one software f64 div per step
+        if (i > t) row[t] *= rdv;
+        else if (i == t) row[t] = dv;
+#pragma unroll
+        for (int j = t + 1; j < 16; ++j) {
+          const double ljt = __shfl(row[t], j, 16);
+          if (i >= j) row[j] = fma(-row[t], ljt, row[j]);
         }
-        if (i == t) A_(k0 + t, k0 + t) = dv;
-        // rank-1 update of the remaining lower part of the block
-        for (int j = t + 1; j <= i; ++j)
-          A_(k0 + i, k0 + j) -= lit * A_(k0 + j, k0 + t);
       }
-      // 2) invert the diagonal block: X = L_kk^-1, column c per lane
-      const int c = lane;
+#pragma unroll
+      for (int c = 0; c < 16; ++c)
+        if (c <= i) A_(k0 + i, k0 + c) = row[c];
+      // 2) invert the block in registers: lane holds column c = i of
+      //    X = L_kk^-1 (forward substitution; all indices compile-time)
+      const int c = i;
+      double diag[16], x[16];
+#pragma unroll
+      for (int r = 0; r < 16; ++r) diag[r] = 1.0 / __shfl(row[r], r, 16);
+#pragma unroll
       for (int r = 0; r < 16; ++r) {
-        double x;
-        if (r < c) x = 0.0;
-        else if (r == c) x = 1.0 / A_(k0 + r, k0 + r);
-        else {
-          double acc = 0.0;
-          for (int t = c; t < r; ++t)
-            acc = fma(A_(k0 + r, k0 + t), inv16[t][c], acc);
-          x = -acc / A_(k0 + r, k0 + r);
+        double acc2 = 0.0;
+#pragma unroll
+        for (int t = 0; t < 16; ++t) {
+          const double lrt = __shfl(row[t], r, 16);
+          if (t >= c && t < r) acc2 = fma(lrt, x[t], acc2);
         }
-        inv16[r][c] = x;
+        x[r] = (r < c) ? 0.0 : (r == c) ? diag[r] : -acc2 * diag[r];
       }
+#pragma unroll
+      for (int r = 0; r < 16; ++r) inv16[r][c] = x[r];
     }
     __syncthreads();
 
     // export invdiag (coalesced)
-    for (int idx = tid; idx < 256; idx += 256) {
+    for (int idx = tid; idx < 256; idx += 512) {
       const int r = idx / 16, c = idx % 16;
       invd[((long)d * nb + kb) * 256 + idx] = inv16[r][c];
     }
 
     // 3) panel TRSM: row tiles below the diagonal, P <- P * inv(L_kk)^T
-    for (int rt = kb + 1 + wv; rt < nb; rt += 4) {
+    for (int rt = kb + 1 + wv; rt < nb; rt += 8) {
       const int r0 = rt * NB;
       f64x4 pacc = {0, 0, 0, 0};
 #pragma unroll
@@ -284,7 +297,7 @@ extern "C" __global__ __launch_bounds__(256) void chol_batch_kernel(
     // 4) trailing update: tiles (ib, jb), kb < jb <= ib < nb
     const int t = nb - kb - 1;
     const int ntile = t * (t + 1) / 2;
-    for (int q = wv; q < ntile; q += 4) {
+    for (int q = wv; q < ntile; q += 8) {
       // triangular index -> (ib, jb), row-major over the lower wedge
       int ib = kb + 1, rem = q;
       while (rem > ib - kb - 1) { rem -= (ib - kb); ++ib; }
@@ -308,7 +321,7 @@ extern "C" __global__ __launch_bounds__(256) void chol_batch_kernel(
   }
 
   // write back L (full rows; upper-triangle junk is never read)
-  for (int idx = tid; idx < mp * mp; idx += 256) {
+  for (int idx = tid; idx < mp * mp; idx += 512) {
     const int i = idx / mp, j = idx % mp;
     L[((long)d * mp + i) * mp + j] = A_(i, j);
   }
@@ -318,128 +331,159 @@ extern "C" __global__ __launch_bounds__(256) void chol_batch_kernel(
 // trsm_fp: per (draw, frequency-tile) forward-substitute
 //   W = L^-1 [B_cols | u]   (u = the T^T N^-1 r column, solved
 //   redundantly per tile: one column vs 126)
-// entirely in LDS, then the fused per-frequency reduction
+// ENTIRELY IN REGISTERS, fused with the per-frequency reduction
 //   M = sNs - [Ws.Ws, Ws.Wc; ., Wc.Wc],  N = sNr - [Ws.wu, Wc.wu]
 //   Fp[d,f] += 0.5 * N^T M^-1 N   (closed-form 2x2)
+//
+// Key identity (probed layout, tools/mfma_probe.hip): the MFMA D/C
+// accumulator layout of a 16x16 tile IS the B-fragment layout across
+// k-steps -- value (row 4v+g, col j) lives in lane (g<<4)|j reg v, and
+// the b-operand for k-step kk wants (row kk*4 + (l>>4), col l&15),
+// i.e. reg kk of the SAME lane.  So the blocked forward substitution
+// runs with W held in registers (acc layout), zero cross-lane moves:
+//   acc  = -RHS_rb;  acc += sum_cb L[rb,cb] (x) W[cb]   (a from LDS Lp)
+//   W[rb] = Iv[rb] (x) (-acc)                           (a from LDS Iv)
+// LDS holds only the L row-panel, the inverted diagonal blocks, and a
+// 16-row staging buffer for the fused reduction -> ~50 KB, 2 WG/CU,
+// vs 151 KB / 1 WG/CU for the LDS-resident W variant
+// (profiles/r01_initial_stats.md).
 // grid.x = ceil(F / 63), grid.y = D;  block = 512 (8 waves)
-// cols layout in LDS W: [s0 c0 s1 c1 ... s62 c62 | u | pad]
+// cols layout: [s0 c0 s1 c1 ... s62 c62 | u | pad]
 // ---------------------------------------------------------------------
 #define FPT_COLS 128
 #define FPT_FREQS 63
-#define W_(r, c) Wsh[(r) * (FPT_COLS + 1) + (c)]
+#define NBMAX (FASTFP_MAXMP / 16)
 
-extern "C" __global__ __launch_bounds__(512) void trsm_fp_kernel(
+template <int NBT>
+__global__ __launch_bounds__(512, 4) void trsm_fp_kernel(
     const double* __restrict__ L /*(D,mp,mp)*/,
     const double* __restrict__ invd /*(D, mp/16, 16, 16)*/,
     const double* __restrict__ RHS /*(mp, 2F+1)*/,
     const double* __restrict__ sNs /*(3,F)*/,
-    const double* __restrict__ sNr /*(2,F)*/, int mp, int F, int D,
+    const double* __restrict__ sNr /*(2,F)*/, int F, int D,
     double* __restrict__ fp /*(D,F)*/) {
-  __shared__ double Wsh[FASTFP_MAXMP * (FPT_COLS + 1)];
+  // NBT = mp/16 is a template parameter so every W[] index below is
+  // compile-time: with a runtime index the register array is demoted to
+  // scratch (288 B/lane measured) and every MFMA b-operand becomes a
+  // memory load -- the whole point of the register-resident design.
+  constexpr int mp = NBT * 16;
   __shared__ double Lp[16][FASTFP_MAXMP + 1];
-  __shared__ double Iv[16][17];
+  __shared__ double Iv[NBMAX][16][17];
+  __shared__ double Wred[16][FPT_COLS + 1];
 
   const int d = blockIdx.y;
   const int f0 = blockIdx.x * FPT_FREQS;
   const int tid = threadIdx.x;
   const int lane = tid & 63;
   const int wv = tid >> 6;
-  const int nb = mp >> 4;
   const long ldr = 2L * F + 1;
   const double* Ld = L + (long)d * mp * mp;
-  const double* Ivd = invd + (long)d * nb * 256;
+  const double* Ivd = invd + (long)d * NBT * 256;
+  const int jw = wv * 16;      // this wave's 16-column strip
+  const int li = lane & 15;    // a-frag row / strip column
+  const int lk = lane >> 4;    // k index / acc row group
 
-  // stage RHS columns [2*f0 .. 2*f0+125] + u into W (zero-fill tail)
-  for (int idx = tid; idx < mp * 126; idx += 512) {
-    const int r = idx / 126, c = idx % 126;
-    const int gc = 2 * f0 + c;
-    W_(r, c) = (gc < 2 * F) ? RHS[r * ldr + gc] : 0.0;
-  }
-  for (int r = tid; r < mp; r += 512) {
-    W_(r, 126) = RHS[r * ldr + 2 * F];
-    W_(r, 127) = 0.0;
-  }
-  __syncthreads();
+  // stage ALL inverted diagonal blocks once
+  for (int idx = tid; idx < NBT * 256; idx += 512)
+    Iv[idx >> 8][(idx >> 4) & 15][idx & 15] = Ivd[idx];
 
-  // blocked forward substitution, row-tile rb at a time
-  const int jw = wv * 16;  // this wave's column strip
-  for (int rb = 0; rb < nb; ++rb) {
-    const int r0 = rb * NB;
-    // stage L row-panel (cols 0..r0) and the inverted diag block
-    for (int idx = tid; idx < 16 * (r0 > 0 ? r0 : 1); idx += 512) {
-      if (r0 == 0) break;
-      const int r = idx / r0, c = idx % r0;
-      Lp[r][c] = Ld[(long)(r0 + r) * mp + c];
-    }
-    for (int idx = tid; idx < 256; idx += 512) {
-      Iv[idx / 16][idx % 16] = Ivd[rb * 256 + idx];
-    }
-    __syncthreads();
-
-    // acc = sum_cb L[rb,cb] W[cb]  -  RHS_rb
-    f64x4 acc;
+  // load this wave's RHS strip into registers (acc layout):
+  // W[rt][v] = RHS[row = rt*16 + 4v + lk][block-col jw + li]
+  const int bc = jw + li;
+  f64x4 W[NBT];
 #pragma unroll
-    for (int v = 0; v < 4; ++v)
-      acc[v] = -W_(r0 + 4 * v + (lane >> 4), jw + (lane & 15));
+  for (int rt = 0; rt < NBT; ++rt) {
+#pragma unroll
+    for (int v = 0; v < 4; ++v) {
+      const long row = rt * 16 + 4 * v + lk;
+      double val = 0.0;
+      if (bc < 126) {
+        const long gc = 2L * f0 + bc;
+        if (gc < 2L * F) val = RHS[row * ldr + gc];
+      } else if (bc == 126) {
+        val = RHS[row * ldr + 2L * F];  // the u column
+      }
+      W[rt][v] = val;
+    }
+  }
+  __syncthreads();  // Iv staged
+
+  // blocked forward substitution, W in registers
+#pragma unroll
+  for (int rb = 0; rb < NBT; ++rb) {
+    if (rb > 0) {
+      __syncthreads();  // prior rb's Lp reads complete
+      const int ncols = rb * 16;
+      for (int idx = tid; idx < 16 * ncols; idx += 512) {
+        const int r = idx / ncols, c = idx % ncols;
+        Lp[r][c] = Ld[(long)(rb * 16 + r) * mp + c];
+      }
+      __syncthreads();
+    }
+    f64x4 acc = -W[rb];  // -RHS_rb
+#pragma unroll
     for (int cb = 0; cb < rb; ++cb) {
 #pragma unroll
       for (int kk = 0; kk < 4; ++kk) {
-        const double a = Lp[lane & 15][cb * 16 + kk * 4 + (lane >> 4)];
-        const double b = W_(cb * 16 + kk * 4 + (lane >> 4), jw + (lane & 15));
-        acc = MFMA_F64(a, b, acc);
+        const double a = Lp[li][cb * 16 + kk * 4 + lk];
+        acc = MFMA_F64(a, W[cb][kk], acc);  // b-frag = reg kk (identity)
       }
     }
-    __syncthreads();  // everyone done READING W[rb] (acc init) before overwrite
-#pragma unroll
-    for (int v = 0; v < 4; ++v)
-      W_(r0 + 4 * v + (lane >> 4), jw + (lane & 15)) = acc[v];
-    __syncthreads();
-    // W[rb] <- inv(L_rb,rb) * (RHS - sum) = Iv * (-Wtmp)
+    // W[rb] = Iv[rb] * (RHS - sum) = Iv[rb] * (-acc)
     f64x4 sol = {0, 0, 0, 0};
 #pragma unroll
     for (int kk = 0; kk < 4; ++kk) {
-      const double a = Iv[lane & 15][kk * 4 + (lane >> 4)];
-      const double b = -W_(r0 + kk * 4 + (lane >> 4), jw + (lane & 15));
-      sol = MFMA_F64(a, b, sol);
+      const double a = Iv[rb][li][kk * 4 + lk];
+      sol = MFMA_F64(a, -acc[kk], sol);
     }
-    __syncthreads();
-#pragma unroll
-    for (int v = 0; v < 4; ++v)
-      W_(r0 + 4 * v + (lane >> 4), jw + (lane & 15)) = sol[v];
-    __syncthreads();
+    W[rb] = sol;
   }
+  __syncthreads();
 
-  // fused per-frequency 2x2 reduction; wave w owns freqs w*8 .. w*8+7
-  for (int q = wv * 8; q < wv * 8 + 8; ++q) {
-    if (q >= FPT_FREQS || f0 + q >= F) continue;
-    const int cs = 2 * q, cc_ = 2 * q + 1;
-    double gss = 0, gcc = 0, gsc = 0, gsu = 0, gcu = 0;
+  // fused reduction: stage one 16-row tile at a time, accumulate the
+  // five per-frequency dots.  Lane (segment lk, row li) accumulates
+  // freq q = wv*8 + pass*4 + lk over rows li, 16+li, 32+li, ...
+  double pss[2] = {0, 0}, pcc[2] = {0, 0}, psc[2] = {0, 0},
+         psu[2] = {0, 0}, pcu[2] = {0, 0};
 #pragma unroll
-    for (int h = 0; h < FASTFP_MAXMP / 64; ++h) {
-      const int r = h * 64 + lane;
-      if (r < mp) {
-        const double ws = W_(r, cs);
-        const double wc = W_(r, cc_);
-        const double wu = W_(r, 126);
-        gss = fma(ws, ws, gss);
-        gcc = fma(wc, wc, gcc);
-        gsc = fma(ws, wc, gsc);
-        gsu = fma(ws, wu, gsu);
-        gcu = fma(wc, wu, gcu);
+  for (int rt = 0; rt < NBT; ++rt) {
+#pragma unroll
+    for (int v = 0; v < 4; ++v) Wred[4 * v + lk][jw + li] = W[rt][v];
+    __syncthreads();
+#pragma unroll
+    for (int pass = 0; pass < 2; ++pass) {
+      const int q = wv * 8 + pass * 4 + lk;
+      if (q < FPT_FREQS && f0 + q < F) {
+        const double ws = Wred[li][2 * q];
+        const double wc = Wred[li][2 * q + 1];
+        const double wu = Wred[li][126];
+        pss[pass] = fma(ws, ws, pss[pass]);
+        pcc[pass] = fma(wc, wc, pcc[pass]);
+        psc[pass] = fma(ws, wc, psc[pass]);
+        psu[pass] = fma(ws, wu, psu[pass]);
+        pcu[pass] = fma(wc, wu, pcu[pass]);
       }
     }
-    gss = wave_reduce_sum(gss);
-    gcc = wave_reduce_sum(gcc);
-    gsc = wave_reduce_sum(gsc);
-    gsu = wave_reduce_sum(gsu);
-    gcu = wave_reduce_sum(gcu);
-    if (lane == 0) {
+    __syncthreads();
+  }
+#pragma unroll
+  for (int pass = 0; pass < 2; ++pass) {
+#pragma unroll
+    for (int off = 8; off > 0; off >>= 1) {
+      pss[pass] += __shfl_down(pss[pass], off, 16);
+      pcc[pass] += __shfl_down(pcc[pass], off, 16);
+      psc[pass] += __shfl_down(psc[pass], off, 16);
+      psu[pass] += __shfl_down(psu[pass], off, 16);
+      pcu[pass] += __shfl_down(pcu[pass], off, 16);
+    }
+    const int q = wv * 8 + pass * 4 + lk;
+    if (li == 0 && q < FPT_FREQS && f0 + q < F) {
       const int f = f0 + q;
-      const double M11 = sNs[f] - gss;
-      const double M22 = sNs[F + f] - gcc;
-      const double M12 = sNs[2 * F + f] - gsc;
-      const double N1 = sNr[f] - gsu;
-      const double N2 = sNr[F + f] - gcu;
+      const double M11 = sNs[f] - pss[pass];
+      const double M22 = sNs[F + f] - pcc[pass];
+      const double M12 = sNs[2 * F + f] - psc[pass];
+      const double N1 = sNr[f] - psu[pass];
+      const double N2 = sNr[F + f] - pcu[pass];
       const double det = fma(M11, M22, -M12 * M12);
       const double num =
           fma(N1 * N1, M22, fma(-2.0 * N1, N2 * M12, N2 * N2 * M11));
@@ -472,7 +516,7 @@ void launch_sbgemm(const double* T, const double* toas, const double* ninv,
 
 void launch_chol_batch(const double* TNT, const double* phiinv, int m, int mp,
                        int D, double* L, double* invd, hipStream_t stream) {
-  hipLaunchKernelGGL(chol_batch_kernel, dim3(D), dim3(256), 0, stream, TNT,
+  hipLaunchKernelGGL(chol_batch_kernel, dim3(D), dim3(512), 0, stream, TNT,
                      phiinv, m, mp, D, L, invd);
 }
 
@@ -480,8 +524,15 @@ void launch_trsm_fp(const double* L, const double* invd, const double* RHS,
                     const double* sNs, const double* sNr, int mp, int F,
                     int D, double* fp, hipStream_t stream) {
   const int ftiles = (F + FPT_FREQS - 1) / FPT_FREQS;
-  hipLaunchKernelGGL(trsm_fp_kernel, dim3(ftiles, D), dim3(512), 0, stream,
-                     L, invd, RHS, sNs, sNr, mp, F, D, fp);
+  const dim3 grid(ftiles, D), blk(512);
+  switch (mp >> 4) {
+#define TRSM_CASE(NBT) \
+    case NBT: hipLaunchKernelGGL(trsm_fp_kernel<NBT>, grid, blk, 0, stream, \
+                                 L, invd, RHS, sNs, sNr, F, D, fp); break;
+    TRSM_CASE(1) TRSM_CASE(2) TRSM_CASE(3) TRSM_CASE(4)
+    TRSM_CASE(5) TRSM_CASE(6) TRSM_CASE(7) TRSM_CASE(8)
+#undef TRSM_CASE
+  }
 }
 
 }  // extern "C"
