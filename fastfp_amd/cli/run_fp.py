@@ -39,6 +39,7 @@ def main(
     device=None,
     rn_comps=30,
     gwb_comps=30,
+    ecorr_kernel=False,
 ):
     logging.basicConfig(format="%(levelname)s: %(message)s", level=logging.INFO)
     logger = logging.getLogger(__name__)
@@ -58,7 +59,8 @@ def main(
     noise["gw_log10_A"] = float(np.log10(2e-15))
 
     pta = initialize_pta(
-        psrs, noise, inc_cp=True, rn_comps=rn_comps, gwb_comps=gwb_comps
+        psrs, noise, inc_cp=True, rn_comps=rn_comps, gwb_comps=gwb_comps,
+        ecorr_kernel=ecorr_kernel,
     )
 
     t0 = time.perf_counter()
@@ -94,6 +96,8 @@ def cli():
     parser.add_argument("--device", type=str, default=None)
     parser.add_argument("--rn_comps", type=int, default=30)
     parser.add_argument("--gwb_comps", type=int, default=30)
+    parser.add_argument("--ecorr_kernel", action="store_true",
+                        help="model ECORR as block-diagonal white noise")
     main(**vars(parser.parse_args()))
 
 

@@ -79,6 +79,7 @@ def main(
     device=None,
     seed=0,
     resume=False,
+    ecorr_kernel=False,
 ):
     logging.basicConfig(format="%(levelname)s: %(message)s", level=logging.INFO)
     logger = logging.getLogger(__name__)
@@ -111,6 +112,7 @@ def main(
         rn_comps=nrncomps,
         gwb_comps=ngwbcomps,
         inc_ecorr=inc_ecorr,
+        ecorr_kernel=ecorr_kernel,
     )
     nmfp = setup_fp_model(psrs, noise, pta=pta)
 
@@ -192,6 +194,10 @@ def cli():
     parser.add_argument("--seed", type=int, default=0)
     parser.add_argument("--resume", action="store_true",
                         help="skip draw batches already on disk")
+    parser.add_argument("--ecorr_kernel", action="store_true",
+                        help="model ECORR as block-diagonal white noise "
+                             "(EcorrKernelNoise; the reference's "
+                             "unsupported case)")
     main(**vars(parser.parse_args()))
 
 

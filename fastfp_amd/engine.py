@@ -46,19 +46,42 @@ def _t64(x, device):
 
 
 class PulsarBlock:
-    """Per-pulsar device tensors + fixed precompute."""
+    """Per-pulsar device tensors + fixed precompute.
+
+    ``Nvec`` may be a per-TOA variance vector (diagonal N) or a
+    :class:`fastfp_amd.blocknoise.BlockNoise` (block-diagonal N, the
+    EcorrKernelNoise case).  For block N the TOAs are re-ordered by the
+    BlockNoise permutation (a pure relabeling — every inner product is
+    permutation-invariant) and the engine precomputes ``V = N^{-1} T``
+    so the frequency GEMM is unchanged: ``T^T N^{-1} S = V^T S``.
+    """
 
     def __init__(self, toas, resid, Nvec, T, device):
+        from fastfp_amd.blocknoise import BlockNoise
+
+        toas = np.asarray(toas, dtype=np.float64)
+        resid = np.asarray(resid, dtype=np.float64)
+        T = np.asarray(T, dtype=np.float64)
+        self.block_noise = Nvec if isinstance(Nvec, BlockNoise) else None
+        if self.block_noise is not None:
+            perm = self.block_noise.perm
+            toas, resid, T = toas[perm], resid[perm], T[perm, :]
         self.toas = _t64(toas, device)
         self.r = _t64(resid, device)
-        self.Nvec = _t64(Nvec, device)
         self.T = _t64(T, device).contiguous()
         self.ntoa, self.m = self.T.shape
 
-        TN = self.T / self.Nvec[:, None]  # N^-1 T  (ntoa, m)
+        if self.block_noise is None:
+            self.Nvec = _t64(Nvec, device)
+            TN = self.T / self.Nvec[:, None]  # N^-1 T  (ntoa, m)
+            self.Nr = self.r / self.Nvec
+        else:
+            self.Nvec = None
+            TN = _t64(self.block_noise.solve(T), device)  # V = N^-1 T
+            self.Nr = _t64(self.block_noise.solve(resid), device)
+        self.V = TN.contiguous()
         self.TNT = self.T.transpose(0, 1) @ TN
         self.TNr = TN.transpose(0, 1) @ self.r
-        self.rNr = torch.dot(self.r, self.r / self.Nvec)
 
         # filled by freq precompute:
         self.RHS = None  # (m, 2F+1): interleaved [s_f, c_f] columns + TNr
@@ -114,33 +137,41 @@ class FpEngine:
         RHS = torch.empty((m, 2 * F + 1), dtype=torch.float64, device=self.device)
         sNs = torch.empty((3, F), dtype=torch.float64, device=self.device)
         sNr = torch.empty((2, F), dtype=torch.float64, device=self.device)
-        Nr = blk.r / blk.Nvec
         for lo in range(0, F, freq_chunk):
             hi = min(lo + freq_chunk, F)
             arg = 2.0 * math.pi * freqs[lo:hi, None] * blk.toas[None, :]  # (Fc, ntoa)
             S = torch.sin(arg)
             C = torch.cos(arg)
-            NS = S / blk.Nvec[None, :]
-            NC = C / blk.Nvec[None, :]
-            # B columns: interleaved sin,cos
-            Bs = NS @ blk.T  # (Fc, m)
-            Bc = NC @ blk.T
+            if blk.block_noise is None:
+                NS = S / blk.Nvec[None, :]
+                NC = C / blk.Nvec[None, :]
+            else:
+                NS = blk.block_noise.solve(S.transpose(0, 1)).transpose(0, 1)
+                NC = blk.block_noise.solve(C.transpose(0, 1)).transpose(0, 1)
+            # B columns: interleaved sin,cos  (B = S N^-1 T = S V)
+            Bs = S @ blk.V  # (Fc, m)
+            Bc = C @ blk.V
             RHS[:, 2 * lo : 2 * hi : 2] = Bs.transpose(0, 1)
             RHS[:, 2 * lo + 1 : 2 * hi : 2] = Bc.transpose(0, 1)
             sNs[0, lo:hi] = (S * NS).sum(dim=1)
             sNs[1, lo:hi] = (C * NC).sum(dim=1)
             sNs[2, lo:hi] = (S * NC).sum(dim=1)
-            sNr[0, lo:hi] = S @ Nr
-            sNr[1, lo:hi] = C @ Nr
+            sNr[0, lo:hi] = S @ blk.Nr
+            sNr[1, lo:hi] = C @ blk.Nr
         RHS[:, -1] = blk.TNr
         blk.RHS, blk.sNs, blk.sNr = RHS, sNs, sNr
 
     def _precompute_hip(self, blk: PulsarBlock, freqs, freq_chunk):
         from fastfp_amd import ops
 
-        blk.RHS, blk.sNs, blk.sNr = ops.freq_precompute(
-            blk.toas, blk.Nvec, blk.r, blk.T, blk.TNr, freqs, freq_chunk
-        )
+        if blk.block_noise is None:
+            blk.RHS, blk.sNs, blk.sNr = ops.freq_precompute(
+                blk.toas, blk.Nvec, blk.r, blk.T, blk.TNr, freqs, freq_chunk
+            )
+        else:
+            blk.RHS, blk.sNs, blk.sNr = ops.freq_precompute_block(
+                blk.toas, blk.block_noise, blk.Nr, blk.V, blk.TNr, freqs
+            )
 
     # ------------------------------------------------------------------
     # sweeps

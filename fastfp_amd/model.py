@@ -31,6 +31,7 @@ from fastfp_amd.bases import (
     timing_model_basis_svd,
 )
 from fastfp_amd.data import get_tspan
+from fastfp_amd.blocknoise import BlockNoise
 from fastfp_amd.noise import (
     CURNContainer,
     GPEcorrContainer,
@@ -57,7 +58,17 @@ class PTAModel:
         simple_wn: bool = True,
         inc_ecorr: bool = False,
         select: str = "backend",
+        ecorr_kernel: bool = False,
     ):
+        """``ecorr_kernel=True`` models ECORR as BLOCK-DIAGONAL white
+        noise (enterprise's EcorrKernelNoise) instead of a basis GP:
+        ``get_ndiag`` then returns :class:`BlockNoise` objects and the
+        basis/phi carry no ECORR columns.  This is the case the
+        reference explicitly does not support
+        (``/root/reference/fastfp/utils.py:30-31``, README.md:22)."""
+        if inc_ecorr and ecorr_kernel:
+            raise ValueError("choose GP ecorr (inc_ecorr) OR kernel ecorr")
+        self.ecorr_kernel = ecorr_kernel
         if inc_cp:
             assert gwb_comps <= rn_comps, (
                 "shared-basis CURN requires gwb_comps <= rn_comps "
@@ -105,9 +116,18 @@ class PTAModel:
             self.ecorr_slices.append(slice(ntm, ntm + n_ec))
             self.rn_slices.append(slice(ntm + n_ec, T.shape[1]))
 
-            self._Nvecs.append(
-                white_noise_nvec(psr, self.noise, simple_wn=simple_wn, select=select)
-            )
+            if ecorr_kernel:
+                self._Nvecs.append(
+                    BlockNoise(
+                        psr, self.noise, simple_wn=simple_wn, select=select
+                    )
+                )
+            else:
+                self._Nvecs.append(
+                    white_noise_nvec(
+                        psr, self.noise, simple_wn=simple_wn, select=select
+                    )
+                )
             self.rn_containers.append(
                 RNContainer(
                     psr,
@@ -144,7 +164,10 @@ class PTAModel:
         return {name: xs[i] for i, name in enumerate(self.param_names)}
 
     def get_ndiag(self, noise: dict = None) -> list:
-        return [nv.copy() for nv in self._Nvecs]
+        return [
+            nv if isinstance(nv, BlockNoise) else nv.copy()
+            for nv in self._Nvecs
+        ]
 
     def get_basis(self, noise: dict = None) -> list:
         return [T.copy() for T in self._Ts]
@@ -161,8 +184,12 @@ class PTAModel:
     def get_TNT(self, noise: dict = None) -> list:
         out = []
         for T, nv in zip(self._Ts, self._Nvecs):
-            TN = T / nv[:, None]
-            out.append(T.T @ TN)
+            if isinstance(nv, BlockNoise):
+                Tp = T[nv.perm, :]
+                out.append(Tp.T @ nv.solve(Tp))
+            else:
+                TN = T / nv[:, None]
+                out.append(T.T @ TN)
         return out
 
 
@@ -175,9 +202,12 @@ def initialize_pta(
     simple_wn=True,
     inc_ecorr=False,
     select="backend",
+    ecorr_kernel=False,
 ) -> PTAModel:
     """Build the PTA model — signature parity with the reference's
-    ``initialize_pta`` (``/root/reference/fastfp/utils.py:104-113``)."""
+    ``initialize_pta`` (``/root/reference/fastfp/utils.py:104-113``),
+    plus ``ecorr_kernel`` for the block-diagonal-N ECORR path the
+    reference lacks."""
     return PTAModel(
         psrs,
         noise,
@@ -187,6 +217,7 @@ def initialize_pta(
         simple_wn=simple_wn,
         inc_ecorr=inc_ecorr,
         select=select,
+        ecorr_kernel=ecorr_kernel,
     )
 
 

@@ -106,6 +106,57 @@ def freq_precompute(toas, Nvec, r, T, TNr, freqs, freq_chunk: int = 8192):
     return RHS, sNs, sNr
 
 
+def freq_precompute_block(toas, block_noise, Nr, V, TNr, freqs):
+    """Block-diagonal-N (EcorrKernelNoise) frequency precompute.
+
+    ``B = T^T N^-1 S = V^T S`` reuses the sbgemm kernel with V in place
+    of T and unit weights; the per-frequency quadratics go through the
+    sigdots_block kernel with the dense per-epoch inverse blocks
+    (factored on device by blockchol_inv).
+    """
+    ext = _try_load()
+    F = int(freqs.shape[0])
+    ntoa, m = V.shape
+    mp = check_m(m)
+    device = V.device
+    freqs = freqs.contiguous()
+
+    if block_noise.max_block > 32:
+        raise NotImplementedError(
+            f"GPU block-noise path supports epochs of <= 32 TOAs "
+            f"(got {block_noise.max_block}); use the CPU engine"
+        )
+    bt = block_noise.tensors(device)
+    # factor + invert the blocks ON DEVICE (block-Cholesky kernel);
+    # validated against the host factorization in tests
+    nvec = torch.as_tensor(block_noise.nvec, dtype=torch.float64, device=device)
+    ecorr2 = torch.as_tensor(
+        block_noise.ecorr2, dtype=torch.float64, device=device
+    )
+    inv_packed, _logdet = ext.blockchol_inv(
+        nvec, ecorr2, bt["offsets"], bt["sizes"], bt["poff"],
+        int(block_noise.inv_packed.shape[0]),
+    )
+    sNs, sNr = ext.sigdots_block(
+        toas.contiguous(), Nr.contiguous(), freqs, inv_packed.contiguous(),
+        bt["offsets"], bt["sizes"], bt["poff"],
+    )
+
+    RHS = torch.zeros((mp, 2 * F + 1), dtype=torch.float64, device=device)
+    F2 = 2 * F
+    ones = torch.ones(ntoa, dtype=torch.float64, device=device)
+    ctiles = (F2 + 63) // 64
+    ksplit = max(1, min(8, (512 + ctiles - 1) // ctiles, (ntoa + 255) // 256))
+    if ksplit == 1:
+        ext.sbgemm(V, toas, ones, freqs, RHS, 0, 2 * F + 1, mp, 1)
+    else:
+        part = torch.empty((ksplit, mp, F2), dtype=torch.float64, device=device)
+        ext.sbgemm(V, toas, ones, freqs, part, mp * F2, F2, mp, ksplit)
+        RHS[:, :F2] = part.sum(dim=0)
+    RHS[:m, -1] = TNr
+    return RHS, sNs, sNr
+
+
 def chol_trsm_fp_accum(TNT, phiinv, RHS, sNs, sNr, fp_out):
     """Batched Cholesky of Sigma = TNT + diag(phiinv) (assembled in
     LDS), then the fused triangular solve of RHS (mp, 2F+1) + 2x2 Fp

@@ -493,6 +493,150 @@ __global__ __launch_bounds__(512, 4) void trsm_fp_kernel(
 }
 
 // ---------------------------------------------------------------------
+// Block-diagonal white noise (EcorrKernelNoise, BASELINE config 4):
+// N is block-diagonal per observing epoch.  The reference never
+// implemented this case (/root/reference/fastfp/utils.py:30-31,
+// README.md:22).  Blocks are contiguous after the BlockNoise TOA
+// permutation (fastfp_amd/blocknoise.py).
+// ---------------------------------------------------------------------
+#define BLK_MAX 32  // max epoch size handled on-GPU (CPU path beyond)
+
+// blockchol_inv: per block, build N_b = diag(nvec_b) + ecorr2_b * J,
+// Cholesky-factor it and write the dense inverse N_b^{-1} = X^T X
+// (X = L^{-1}), packed row-major at poff[b].  One WAVE per block,
+// register-resident rows, cross-lane shfl (width 32) — the same
+// technique as chol_batch's diagonal factor.
+// grid.x = ceil(nblk / 8), block = 512 (8 waves)
+extern "C" __global__ __launch_bounds__(512) void blockchol_inv_kernel(
+    const double* __restrict__ nvec /*(ntoa,) permuted*/,
+    const double* __restrict__ ecorr2 /*(nblk,)*/,
+    const long* __restrict__ offsets, const long* __restrict__ sizes,
+    const long* __restrict__ poff, int nblk,
+    double* __restrict__ inv_packed, double* __restrict__ logdet_blk) {
+  const int b = blockIdx.x * 8 + (threadIdx.x >> 6);
+  if (b >= nblk) return;
+  const int lane = threadIdx.x & 63;
+  const int i = lane & 31;  // row owned by this lane (2x redundant)
+  const int sz = (int)sizes[b];
+  const long o = offsets[b];
+  const double e2 = ecorr2[b];
+
+  double row[BLK_MAX];  // row i of the block, then of L
+#pragma unroll
+  for (int c = 0; c < BLK_MAX; ++c) {
+    double v = (c < sz && i < sz) ? e2 : 0.0;
+    if (c == i && i < sz) v += nvec[o + i];
+    row[c] = v;
+  }
+  // Cholesky, lanes as rows (width-32 shfl)
+  double ld = 0.0;
+#pragma unroll
+  for (int t = 0; t < BLK_MAX; ++t) {
+    if (t >= sz) break;
+    const double att = __shfl(row[t], t, 32);
+    const double dv = sqrt(att);
+    const double rdv = 1.0 / dv;
+    if (i == t) { row[t] = dv; ld += 2.0 * log(dv); }
+    else if (i > t) row[t] *= rdv;
+#pragma unroll
+    for (int j = t + 1; j < BLK_MAX; ++j) {
+      const double ljt = __shfl(row[t], j, 32);
+      if (j < sz && i >= j) row[j] = fma(-row[t], ljt, row[j]);
+    }
+  }
+  // X = L^{-1}: lane holds column c = i
+  const int c = i;
+  double diag[BLK_MAX], x[BLK_MAX];
+#pragma unroll
+  for (int r = 0; r < BLK_MAX; ++r)
+    diag[r] = (r < sz) ? 1.0 / __shfl(row[r], r, 32) : 0.0;
+#pragma unroll
+  for (int r = 0; r < BLK_MAX; ++r) {
+    double acc = 0.0;
+#pragma unroll
+    for (int t = 0; t < BLK_MAX; ++t) {
+      const double lrt = __shfl(row[t], r, 32);
+      if (t >= c && t < r && t < sz) acc = fma(lrt, x[t], acc);
+    }
+    x[r] = (r < c) ? 0.0 : (r == c) ? diag[r] : -acc * diag[r];
+  }
+  // N^{-1}[i2][c] = sum_t X[t][i2] * X[t][c]; lane c writes column c
+  double* out = inv_packed + poff[b];
+#pragma unroll
+  for (int i2 = 0; i2 < BLK_MAX; ++i2) {
+    if (i2 >= sz) break;
+    double acc = 0.0;
+#pragma unroll
+    for (int t = 0; t < BLK_MAX; ++t) {
+      const double xti = __shfl(x[t], i2, 32);
+      if (t < sz) acc = fma(xti, x[t], acc);
+    }
+    if (c < sz && lane < 32) out[(long)i2 * sz + c] = acc;
+  }
+  // logdet: each lane contributed its own diagonal's 2*log(dv)
+  for (int off = 16; off > 0; off >>= 1) ld += __shfl_down(ld, off, 32);
+  if (lane == 0) logdet_blk[b] = ld;
+}
+
+// sigdots_block: the five per-frequency dots with BLOCK-diagonal N.
+// sr/cr use the precomputed nr = N^{-1} r vector; the quadratics use
+// the packed dense block inverses.  grid.x = F, block = 256.
+extern "C" __global__ __launch_bounds__(256) void sigdots_block_kernel(
+    const double* __restrict__ toas, const double* __restrict__ nr,
+    const double* __restrict__ freqs,
+    const double* __restrict__ inv_packed,
+    const long* __restrict__ offsets, const long* __restrict__ sizes,
+    const long* __restrict__ poff, int nblk, int ntoa, int F,
+    double* __restrict__ sNs, double* __restrict__ sNr) {
+  const int f = blockIdx.x;
+  if (f >= F) return;
+  const double w = 2.0 * M_PI * freqs[f];
+  double ss = 0, cc = 0, sc = 0, sr = 0, cr = 0;
+  for (int i = threadIdx.x; i < ntoa; i += blockDim.x) {
+    double s, c;
+    sincos(w * toas[i], &s, &c);
+    sr = fma(s, nr[i], sr);
+    cr = fma(c, nr[i], cr);
+  }
+  for (int b = threadIdx.x; b < nblk; b += blockDim.x) {
+    const int sz = (int)sizes[b];
+    const long o = offsets[b];
+    const double* Binv = inv_packed + poff[b];
+    double sv[BLK_MAX], cv[BLK_MAX];
+    for (int i = 0; i < sz; ++i) sincos(w * toas[o + i], &sv[i], &cv[i]);
+    for (int i = 0; i < sz; ++i) {
+      for (int j = 0; j < sz; ++j) {
+        const double wij = Binv[(long)i * sz + j];
+        ss = fma(sv[i] * wij, sv[j], ss);
+        cc = fma(cv[i] * wij, cv[j], cc);
+        sc = fma(sv[i] * wij, cv[j], sc);
+      }
+    }
+  }
+  __shared__ double red[4][5];
+  ss = wave_reduce_sum(ss);
+  cc = wave_reduce_sum(cc);
+  sc = wave_reduce_sum(sc);
+  sr = wave_reduce_sum(sr);
+  cr = wave_reduce_sum(cr);
+  const int lane = threadIdx.x & 63, wv = threadIdx.x >> 6;
+  if (lane == 0) {
+    red[wv][0] = ss; red[wv][1] = cc; red[wv][2] = sc;
+    red[wv][3] = sr; red[wv][4] = cr;
+  }
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    double a0 = 0, a1 = 0, a2 = 0, a3 = 0, a4 = 0;
+    for (int v = 0; v < 4; ++v) {
+      a0 += red[v][0]; a1 += red[v][1]; a2 += red[v][2];
+      a3 += red[v][3]; a4 += red[v][4];
+    }
+    sNs[f] = a0; sNs[F + f] = a1; sNs[2 * F + f] = a2;
+    sNr[f] = a3; sNr[F + f] = a4;
+  }
+}
+
+// ---------------------------------------------------------------------
 // host-side launchers (called from bindings.cpp)
 // ---------------------------------------------------------------------
 extern "C" {
@@ -533,6 +677,25 @@ void launch_trsm_fp(const double* L, const double* invd, const double* RHS,
     TRSM_CASE(5) TRSM_CASE(6) TRSM_CASE(7) TRSM_CASE(8)
 #undef TRSM_CASE
   }
+}
+
+void launch_blockchol_inv(const double* nvec, const double* ecorr2,
+                          const long* offsets, const long* sizes,
+                          const long* poff, int nblk, double* inv_packed,
+                          double* logdet_blk, hipStream_t stream) {
+  hipLaunchKernelGGL(blockchol_inv_kernel, dim3((nblk + 7) / 8), dim3(512),
+                     0, stream, nvec, ecorr2, offsets, sizes, poff, nblk,
+                     inv_packed, logdet_blk);
+}
+
+void launch_sigdots_block(const double* toas, const double* nr,
+                          const double* freqs, const double* inv_packed,
+                          const long* offsets, const long* sizes,
+                          const long* poff, int nblk, int ntoa, int F,
+                          double* sNs, double* sNr, hipStream_t stream) {
+  hipLaunchKernelGGL(sigdots_block_kernel, dim3(F), dim3(256), 0, stream,
+                     toas, nr, freqs, inv_packed, offsets, sizes, poff,
+                     nblk, ntoa, F, sNs, sNr);
 }
 
 }  // extern "C"

@@ -17,9 +17,26 @@ import torch
 
 
 def get_xCy(Nvec, T, sigma, x, y):
-    """Compute ``x^T C^-1 y`` for diagonal N.  Accepts numpy arrays or
+    """Compute ``x^T C^-1 y``.  ``Nvec`` is a diagonal-variance vector,
+    or a :class:`fastfp_amd.blocknoise.BlockNoise` for block-diagonal N
+    (the EcorrKernelNoise case the reference does not support,
+    ``/root/reference/fastfp/utils.py:30-31``).  Accepts numpy arrays or
     torch tensors; returns a python float (numpy inputs) or 0-dim torch
     tensor (torch inputs)."""
+    from fastfp_amd.blocknoise import BlockNoise
+
+    if isinstance(Nvec, BlockNoise):
+        perm = Nvec.perm
+        xp = np.asarray(x, dtype=np.float64)[perm]
+        yp = np.asarray(y, dtype=np.float64)[perm]
+        Tp = np.asarray(T, dtype=np.float64)[perm, :]
+        Nx = Nvec.solve(xp)
+        Ny = Nvec.solve(yp)
+        TNx = Tp.T @ Nx
+        TNy = Tp.T @ Ny
+        xNy = float(xp @ Ny)
+        return xNy - float(TNx @ np.linalg.solve(np.asarray(sigma), TNy))
+
     torch_in = any(isinstance(a, torch.Tensor) for a in (Nvec, T, sigma, x, y))
 
     def cv(a):
