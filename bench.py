@@ -58,6 +58,12 @@ def build_problem(args, device, rank):
     eng = FpEngine(psrs, Nvecs, Ts, device=device)
     freqs = np.arange(1, args.freqs + 1) / pta.Tspan
     eng.precompute(freqs, freq_chunk=args.freq_chunk)
+    if not args.no_compress:
+        probe = {k: v for k, v in noise.items() if k in pta.params}
+        eng.enable_draw_compression(
+            [c.var_slice for c in pta.rn_containers],
+            [c.get_phiinv(probe).to(device) for c in pta.rn_containers],
+        )
 
     for cont in pta.rn_containers:
         cont.to(device)
@@ -102,6 +108,8 @@ def main():
     ap.add_argument("--device", type=str, default=None)
     ap.add_argument("--no-graph", action="store_true",
                     help="disable hipGraph capture of the step")
+    ap.add_argument("--no-compress", action="store_true",
+                    help="disable the Schur draw compression")
     args = ap.parse_args()
 
     rank, world, device = init_distributed(

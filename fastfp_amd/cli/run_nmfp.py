@@ -147,6 +147,12 @@ def main(
 
     eng = FpEngine(psrs, Nvecs, Ts, device=dev)
     eng.precompute(freqs)
+    # Schur draw compression: per-draw solves run at the variable-bin
+    # dimension only (docs/DESIGN.md)
+    eng.enable_draw_compression(
+        [c.var_slice for c in pta.rn_containers],
+        [c.get_phiinv(noise) for c in pta.rn_containers],
+    )
 
     t0 = time.perf_counter()
     parts = []

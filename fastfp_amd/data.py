@@ -188,8 +188,13 @@ def _design_matrix(toas: np.ndarray, ntm: int, rng: np.random.Generator) -> np.n
     cols.append(np.cos(2 * np.pi * t0 / yr))
     k = 1
     while len(cols) < ntm:
-        # low-frequency smooth columns; pairs of sin/cos at k/tspan
-        cols.append(np.sin(2 * np.pi * k * t0 / tspan + rng.uniform(0, 2 * np.pi)))
+        # low-frequency smooth columns (DM/FD/jump proxies).  The
+        # half-integer frequencies overlap the red-noise Fourier basis
+        # REALISTICALLY (strong correlation) without being exactly
+        # degenerate with it.
+        cols.append(
+            np.sin(2 * np.pi * (k + 0.37) * t0 / tspan + rng.uniform(0, 2 * np.pi))
+        )
         k += 1
     return np.stack(cols[:ntm], axis=1)
 

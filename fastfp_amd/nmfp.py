@@ -79,6 +79,7 @@ class NMFp:
         draw_chunk: int = 32,
         freq_chunk: int = 2048,
         engine: FpEngine = None,
+        compress: bool = True,
     ) -> np.ndarray:
         """NM-Fp over (draws x freqs).
 
@@ -92,8 +93,20 @@ class NMFp:
         if engine is None:
             engine = FpEngine(self.psrs, Nvecs, Ts, device=device)
             engine.precompute(freqs, freq_chunk=freq_chunk)
+            if compress:
+                # fixed phi^-1 from any parameter point; variable bins
+                # are zeroed inside enable_draw_compression
+                probe = {k: (v[0] if np.ndim(v) else v) for k, v in samples.items()}
+                engine.enable_draw_compression(
+                    [sig.var_slice for sig in self.rn_sigs],
+                    [sig.get_phiinv(probe) for sig in self.rn_sigs],
+                )
         phiinvs = [sig.get_phiinv(samples) for sig in self.rn_sigs]
         # scalar-parameter dicts produce (m,) vectors; promote to (1, m)
         phiinvs = [p[None, :] if p.dim() == 1 else p for p in phiinvs]
+        if compress and engine.compression_margin(phiinvs) < 1e3:
+            # parameter draws too close to the Sigma_0 jitter floor:
+            # fall back to the exact direct path
+            engine.disable_draw_compression()
         fp = engine.sweep(phiinvs=phiinvs, draw_chunk=draw_chunk)
         return fp.cpu().numpy()

@@ -307,6 +307,18 @@ class RNContainer:
         return self.update_phi(pars)
 
     @property
+    def var_slice(self) -> slice:
+        """Basis columns whose prior VARIES with the sampled parameters
+        (the red-noise(+CURN) Fourier bins) — the draw-compression
+        dimension (docs/DESIGN.md)."""
+        start = 0
+        if self.inc_tm:
+            start += self.tm_weights.shape[0]
+        if self.gp_ecorr:
+            start += self.ecorr_container.get_phi().shape[0]
+        return slice(start, self.nphi)
+
+    @property
     def nphi(self) -> int:
         n = self.Ffreqs.shape[0]
         if self.inc_tm:

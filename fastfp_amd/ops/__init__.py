@@ -157,12 +157,15 @@ def freq_precompute_block(toas, block_noise, Nr, V, TNr, freqs):
     return RHS, sNs, sNr
 
 
-def chol_trsm_fp_accum(TNT, phiinv, RHS, sNs, sNr, fp_out):
+def chol_trsm_fp_accum(TNT, phiinv, RHS, sNs, sNr, fp_out, gsign=1.0):
     """Batched Cholesky of Sigma = TNT + diag(phiinv) (assembled in
     LDS), then the fused triangular solve of RHS (mp, 2F+1) + 2x2 Fp
-    reduction, accumulating into fp_out (D, F)."""
+    reduction, accumulating into fp_out (D, F).
+
+    ``gsign``: +1 direct path (M = sNs - W.W); -1 Schur-compressed draw
+    path (M = M0 + W.W) — docs/DESIGN.md §draw compression."""
     ext = _try_load()
     m = TNT.shape[0]
     mp = check_m(m)
     L, invd = ext.chol_batch(TNT.contiguous(), phiinv.contiguous(), mp)
-    ext.trsm_fp_accum(L, invd, RHS, sNs, sNr, fp_out)
+    ext.trsm_fp_accum(L, invd, RHS, sNs, sNr, fp_out, gsign)

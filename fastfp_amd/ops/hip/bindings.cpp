@@ -16,8 +16,8 @@ void launch_sbgemm(const double*, const double*, const double*,
 void launch_chol_batch(const double*, const double*, int, int, int, double*,
                        double*, hipStream_t);
 void launch_trsm_fp(const double*, const double*, const double*,
-                    const double*, const double*, int, int, int, double*,
-                    hipStream_t);
+                    const double*, const double*, int, int, int, double,
+                    double*, hipStream_t);
 void launch_blockchol_inv(const double*, const double*, const long*,
                           const long*, const long*, int, double*, double*,
                           hipStream_t);
@@ -101,7 +101,8 @@ std::vector<torch::Tensor> chol_batch(torch::Tensor TNT, torch::Tensor phiinv,
 // trsm_fp_accum: L (D,mp,mp), invd, RHS (mp, 2F+1), sNs (3,F), sNr (2,F),
 // fp (D,F) accumulated in place.
 void trsm_fp_accum(torch::Tensor L, torch::Tensor invd, torch::Tensor RHS,
-                   torch::Tensor sNs, torch::Tensor sNr, torch::Tensor fp) {
+                   torch::Tensor sNs, torch::Tensor sNr, torch::Tensor fp,
+                   double gsign) {
   check_f64(L, "L");
   check_f64(invd, "invd");
   check_f64(RHS, "RHS");
@@ -116,8 +117,8 @@ void trsm_fp_accum(torch::Tensor L, torch::Tensor invd, torch::Tensor RHS,
   TORCH_CHECK(mp % 16 == 0 && mp <= 128, "mp must be <=128, multiple of 16");
   launch_trsm_fp(L.data_ptr<double>(), invd.data_ptr<double>(),
                  RHS.data_ptr<double>(), sNs.data_ptr<double>(),
-                 sNr.data_ptr<double>(), mp, F, D, fp.data_ptr<double>(),
-                 stream());
+                 sNr.data_ptr<double>(), mp, F, D, gsign,
+                 fp.data_ptr<double>(), stream());
 }
 
 // blockchol_inv: factor+invert the per-epoch noise blocks on device.

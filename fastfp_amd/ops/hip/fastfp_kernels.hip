@@ -198,14 +198,20 @@ extern "C" __global__ __launch_bounds__(256) void sbgemm_kernel(
 // (width 16) -- the LDS read-modify-write chains of the naive version
 // were the kernel's dominant cost (profiles/r01_initial_stats.md).
 // ---------------------------------------------------------------------
-#define A_(i, j) Ash[(i) * (FASTFP_MAXMP + 1) + (j)]
+#define A_(i, j) Ash[(i) * (NBT * 16 + 1) + (j)]
 
-extern "C" __global__ __launch_bounds__(512) void chol_batch_kernel(
+template <int NBT>
+__global__ __launch_bounds__(512) void chol_batch_kernel(
     const double* __restrict__ TNT /*(m,m)*/,
-    const double* __restrict__ phiinv /*(D,m)*/, int m, int mp, int D,
+    const double* __restrict__ phiinv /*(D,m)*/, int m, int D,
     double* __restrict__ L /*(D,mp,mp)*/,
     double* __restrict__ invd /*(D, mp/16, 16, 16)*/) {
-  __shared__ double Ash[FASTFP_MAXMP * (FASTFP_MAXMP + 1)];
+  // LDS is sized by the template so small matrices keep multiple
+  // workgroups per CU (nb=4: 33 KB -> 4 WG/CU vs one at 134 KB; the
+  // serial diagonal phases then overlap ACROSS workgroups -- the PMC
+  // profile showed 84% of wave cycles parked, profiles/).
+  constexpr int mp = NBT * 16;
+  __shared__ double Ash[mp * (mp + 1)];
   __shared__ double inv16[16][17];
 
   const int d = blockIdx.x;
@@ -213,7 +219,7 @@ extern "C" __global__ __launch_bounds__(512) void chol_batch_kernel(
   const int tid = threadIdx.x;
   const int lane = tid & 63;
   const int wv = tid >> 6;
-  const int nb = mp >> 4;
+  constexpr int nb = NBT;
 
   // assemble Sigma in LDS
   for (int idx = tid; idx < mp * mp; idx += 512) {
@@ -361,7 +367,9 @@ __global__ __launch_bounds__(512, 4) void trsm_fp_kernel(
     const double* __restrict__ RHS /*(mp, 2F+1)*/,
     const double* __restrict__ sNs /*(3,F)*/,
     const double* __restrict__ sNr /*(2,F)*/, int F, int D,
-    double* __restrict__ fp /*(D,F)*/) {
+    double gsign, double* __restrict__ fp /*(D,F)*/) {
+  // gsign: +1 for the direct path (M = sNs - W.W), -1 for the
+  // Schur-compressed draw path (M = M0 + W.W) -- docs/DESIGN.md.
   // NBT = mp/16 is a template parameter so every W[] index below is
   // compile-time: with a runtime index the register array is demoted to
   // scratch (288 B/lane measured) and every MFMA b-operand becomes a
@@ -479,11 +487,11 @@ __global__ __launch_bounds__(512, 4) void trsm_fp_kernel(
     const int q = wv * 8 + pass * 4 + lk;
     if (li == 0 && q < FPT_FREQS && f0 + q < F) {
       const int f = f0 + q;
-      const double M11 = sNs[f] - pss[pass];
-      const double M22 = sNs[F + f] - pcc[pass];
-      const double M12 = sNs[2 * F + f] - psc[pass];
-      const double N1 = sNr[f] - psu[pass];
-      const double N2 = sNr[F + f] - pcu[pass];
+      const double M11 = sNs[f] - gsign * pss[pass];
+      const double M22 = sNs[F + f] - gsign * pcc[pass];
+      const double M12 = sNs[2 * F + f] - gsign * psc[pass];
+      const double N1 = sNr[f] - gsign * psu[pass];
+      const double N2 = sNr[F + f] - gsign * pcu[pass];
       const double det = fma(M11, M22, -M12 * M12);
       const double num =
           fma(N1 * N1, M22, fma(-2.0 * N1, N2 * M12, N2 * N2 * M11));
@@ -660,19 +668,26 @@ void launch_sbgemm(const double* T, const double* toas, const double* ninv,
 
 void launch_chol_batch(const double* TNT, const double* phiinv, int m, int mp,
                        int D, double* L, double* invd, hipStream_t stream) {
-  hipLaunchKernelGGL(chol_batch_kernel, dim3(D), dim3(512), 0, stream, TNT,
-                     phiinv, m, mp, D, L, invd);
+  const dim3 grid(D), blk(512);
+  switch (mp >> 4) {
+#define CHOL_CASE(NBT) \
+    case NBT: hipLaunchKernelGGL(chol_batch_kernel<NBT>, grid, blk, 0, \
+                                 stream, TNT, phiinv, m, D, L, invd); break;
+    CHOL_CASE(1) CHOL_CASE(2) CHOL_CASE(3) CHOL_CASE(4)
+    CHOL_CASE(5) CHOL_CASE(6) CHOL_CASE(7) CHOL_CASE(8)
+#undef CHOL_CASE
+  }
 }
 
 void launch_trsm_fp(const double* L, const double* invd, const double* RHS,
                     const double* sNs, const double* sNr, int mp, int F,
-                    int D, double* fp, hipStream_t stream) {
+                    int D, double gsign, double* fp, hipStream_t stream) {
   const int ftiles = (F + FPT_FREQS - 1) / FPT_FREQS;
   const dim3 grid(ftiles, D), blk(512);
   switch (mp >> 4) {
 #define TRSM_CASE(NBT) \
     case NBT: hipLaunchKernelGGL(trsm_fp_kernel<NBT>, grid, blk, 0, stream, \
-                                 L, invd, RHS, sNs, sNr, F, D, fp); break;
+                                 L, invd, RHS, sNs, sNr, F, D, gsign, fp); break;
     TRSM_CASE(1) TRSM_CASE(2) TRSM_CASE(3) TRSM_CASE(4)
     TRSM_CASE(5) TRSM_CASE(6) TRSM_CASE(7) TRSM_CASE(8)
 #undef TRSM_CASE
