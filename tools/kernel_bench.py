@@ -61,7 +61,7 @@ def main():
 
         L, invd = ext.chol_batch(TNT, phiinv, mp)
         fp = torch.zeros((D, F), dtype=torch.float64, device=DEV)
-        t = timeit(lambda: ext.trsm_fp_accum(L, invd, RHS, sNs, sNr, fp))
+        t = timeit(lambda: ext.trsm_fp_accum(L, invd, RHS, sNs, sNr, fp, 1.0))
         flops = D * (mp * mp / 2) * (2 * F + 2) * 2  # TRSM MACs*2
         print(f"trsm  D={D:5d}: {t*1e6:9.1f} us  {flops/t/1e12:6.2f} TF/s"
               f"  ({flops:.3g} flops)")
