@@ -111,11 +111,15 @@ extern "C" __global__ __launch_bounds__(256) void sigdots_kernel(
 // planes (deterministic torch reduction afterwards).
 // grid.x = ceil(2F / 64) col tiles, grid.y = ksplit;  block = 256 (4 waves)
 // ---------------------------------------------------------------------
-extern "C" __global__ __launch_bounds__(256) void sbgemm_kernel(
+extern "C" __global__ __launch_bounds__(512) void sbgemm_kernel(
     const double* __restrict__ T /*(ntoa, m) row-major*/,
     const double* __restrict__ toas, const double* __restrict__ ninv,
     const double* __restrict__ freqs, int ntoa, int m, int mp, int F2,
     double* __restrict__ out, long plane_stride, long ldo) {
+  // 8 waves: 4 column strips x 2 row halves.  The 4-wave variant held
+  // 8 accumulator tiles per wave (64+ AGPR) which capped occupancy at
+  // 3 waves/SIMD; splitting the M dim halves the accumulator and lets
+  // more waves cover the sincos + LDS latency.
   __shared__ double lT[16][FASTFP_MAXMP + 1];
   __shared__ double lS[64][17];
 
@@ -123,7 +127,13 @@ extern "C" __global__ __launch_bounds__(256) void sbgemm_kernel(
   const int lane = tid & 63;
   const int wv = tid >> 6;
   const int c0 = blockIdx.x * 64;       // first output column of this tile
-  const int nrt = mp >> 4;              // row tiles
+  const int jw = (wv & 3) * 16;         // column strip
+  const int rh = wv >> 2;               // row half (0: rows 0-63, 1: 64-127)
+  const int rowbase = rh * 64;
+  const int nrt_tot = mp >> 4;
+  // row tiles this wave owns (half 1 may be empty for small mp)
+  const int rt_lo = min(rh * 4, nrt_tot);
+  const int rt_hi = min(rt_lo + 4, nrt_tot);
 
   // K range of this split
   const int ks = gridDim.y;
@@ -132,58 +142,57 @@ extern "C" __global__ __launch_bounds__(256) void sbgemm_kernel(
   const int kend = min(ntoa, kbeg + kchunk);
   double* outp = out + (long)blockIdx.y * plane_stride;
 
-  f64x4 acc[FASTFP_MAXMP / 16];
+  f64x4 acc[4];
 #pragma unroll
-  for (int rt = 0; rt < FASTFP_MAXMP / 16; ++rt) acc[rt] = f64x4{0, 0, 0, 0};
+  for (int q = 0; q < 4; ++q) acc[q] = f64x4{0, 0, 0, 0};
 
   for (int k0 = kbeg; k0 < kend; k0 += 16) {
     // stage T panel rows k0..k0+15 (zero-padded)
-    for (int idx = tid; idx < 16 * mp; idx += 256) {
+    for (int idx = tid; idx < 16 * mp; idx += 512) {
       const int k = idx / mp, j = idx % mp;
       const int gk = k0 + k;
       lT[k][j] = (gk < kend && j < m) ? T[(long)gk * m + j] : 0.0;
     }
     // stage trig panel: 32 freq-pairs x 16 toas, one sincos each
-    for (int idx = tid; idx < 32 * 16; idx += 256) {
+    for (int idx = tid; idx < 32 * 16; idx += 512) {
       const int p = idx / 16, k = idx % 16;
       const int gk = k0 + k;
       const int ceven = c0 + 2 * p;
-      double s = 0.0, c = 0.0;
+      double sv = 0.0, cv = 0.0;
       if (gk < kend && ceven < F2) {
         const double wf = 2.0 * M_PI * freqs[ceven >> 1];
-        sincos(wf * toas[gk], &s, &c);
+        sincos(wf * toas[gk], &sv, &cv);
         const double ni = ninv[gk];
-        s *= ni; c *= ni;
+        sv *= ni; cv *= ni;
       }
-      lS[2 * p][k] = s;
-      if (2 * p + 1 < 64) lS[2 * p + 1][k] = c;
+      lS[2 * p][k] = sv;
+      lS[2 * p + 1][k] = cv;
     }
     __syncthreads();
 
-    const int jw = wv * 16;  // this wave's 16-column strip
 #pragma unroll
     for (int kk = 0; kk < 4; ++kk) {
       const double b = lS[jw + (lane & 15)][kk * 4 + (lane >> 4)];
 #pragma unroll
-      for (int rt = 0; rt < FASTFP_MAXMP / 16; ++rt) {
-        if (rt >= nrt) break;
-        const double a = lT[kk * 4 + (lane >> 4)][rt * 16 + (lane & 15)];
-        acc[rt] = MFMA_F64(a, b, acc[rt]);
+      for (int q = 0; q < 4; ++q) {
+        if (rt_lo + q >= rt_hi) break;
+        const double a =
+            lT[kk * 4 + (lane >> 4)][rowbase + q * 16 + (lane & 15)];
+        acc[q] = MFMA_F64(a, b, acc[q]);
       }
     }
     __syncthreads();
   }
 
   // epilogue: store (row j, col c) — cols contiguous, coalesced
-  const int jw = wv * 16;
 #pragma unroll
-  for (int rt = 0; rt < FASTFP_MAXMP / 16; ++rt) {
-    if (rt >= nrt) break;
+  for (int q = 0; q < 4; ++q) {
+    if (rt_lo + q >= rt_hi) break;
 #pragma unroll
     for (int v = 0; v < 4; ++v) {
-      const int row = rt * 16 + 4 * v + (lane >> 4);
+      const int row = rowbase + q * 16 + 4 * v + (lane >> 4);
       const int col = c0 + jw + (lane & 15);
-      if (col < F2) outp[(long)row * ldo + col] = acc[rt][v];
+      if (col < F2) outp[(long)row * ldo + col] = acc[q][v];
     }
   }
 }
@@ -360,7 +369,7 @@ __global__ __launch_bounds__(512) void chol_batch_kernel(
 #define FPT_FREQS 63
 #define NBMAX (FASTFP_MAXMP / 16)
 
-template <int NBT>
+template <int NBT, int DPG>
 __global__ __launch_bounds__(512, 4) void trsm_fp_kernel(
     const double* __restrict__ L /*(D,mp,mp)*/,
     const double* __restrict__ invd /*(D, mp/16, 16, 16)*/,
@@ -374,31 +383,38 @@ __global__ __launch_bounds__(512, 4) void trsm_fp_kernel(
   // compile-time: with a runtime index the register array is demoted to
   // scratch (288 B/lane measured) and every MFMA b-operand becomes a
   // memory load -- the whole point of the register-resident design.
+  // DPG (1 or 2) = draws per workgroup: at the compressed path's small
+  // NBT the solve is short, so two draws share the Lp/Iv staging and
+  // barriers, and their MFMA chains interleave on the pipe.
   constexpr int mp = NBT * 16;
-  __shared__ double Lp[16][FASTFP_MAXMP + 1];
-  __shared__ double Iv[NBMAX][16][17];
-  __shared__ double Wred[16][FPT_COLS + 1];
+  __shared__ double Lp[DPG][16][NBT * 16 + 1];
+  __shared__ double Iv[DPG][NBT][16][17];
+  __shared__ double Wu[DPG][NBT * 16];  // the solved u column
 
-  const int d = blockIdx.y;
+  const int d0 = blockIdx.y * DPG;
   const int f0 = blockIdx.x * FPT_FREQS;
   const int tid = threadIdx.x;
   const int lane = tid & 63;
   const int wv = tid >> 6;
   const long ldr = 2L * F + 1;
-  const double* Ld = L + (long)d * mp * mp;
-  const double* Ivd = invd + (long)d * NBT * 256;
   const int jw = wv * 16;      // this wave's 16-column strip
   const int li = lane & 15;    // a-frag row / strip column
   const int lk = lane >> 4;    // k index / acc row group
+  const int ndr = min(DPG, D - d0);  // draws handled here (edge: D odd)
 
-  // stage ALL inverted diagonal blocks once
-  for (int idx = tid; idx < NBT * 256; idx += 512)
-    Iv[idx >> 8][(idx >> 4) & 15][idx & 15] = Ivd[idx];
+  // stage ALL inverted diagonal blocks once (both draws)
+  for (int idx = tid; idx < ndr * NBT * 256; idx += 512) {
+    const int e = idx / (NBT * 256);
+    const int r = idx % (NBT * 256);
+    Iv[e][r >> 8][(r >> 4) & 15][r & 15] =
+        invd[((long)(d0 + e) * NBT) * 256 + r];
+  }
 
   // load this wave's RHS strip into registers (acc layout):
-  // W[rt][v] = RHS[row = rt*16 + 4v + lk][block-col jw + li]
+  // W[e][rt][v] = RHS[row = rt*16 + 4v + lk][block-col jw + li]
+  // (same RHS for every draw)
   const int bc = jw + li;
-  f64x4 W[NBT];
+  f64x4 W[DPG][NBT];
 #pragma unroll
   for (int rt = 0; rt < NBT; ++rt) {
 #pragma unroll
@@ -411,91 +427,113 @@ __global__ __launch_bounds__(512, 4) void trsm_fp_kernel(
       } else if (bc == 126) {
         val = RHS[row * ldr + 2L * F];  // the u column
       }
-      W[rt][v] = val;
+#pragma unroll
+      for (int e = 0; e < DPG; ++e) W[e][rt][v] = val;
     }
   }
   __syncthreads();  // Iv staged
 
-  // blocked forward substitution, W in registers
+  // blocked forward substitution, W in registers, DPG draws interleaved
 #pragma unroll
   for (int rb = 0; rb < NBT; ++rb) {
     if (rb > 0) {
       __syncthreads();  // prior rb's Lp reads complete
       const int ncols = rb * 16;
-      for (int idx = tid; idx < 16 * ncols; idx += 512) {
-        const int r = idx / ncols, c = idx % ncols;
-        Lp[r][c] = Ld[(long)(rb * 16 + r) * mp + c];
+      for (int idx = tid; idx < ndr * 16 * ncols; idx += 512) {
+        const int e = idx / (16 * ncols);
+        const int q = idx % (16 * ncols);
+        const int r = q / ncols, c = q % ncols;
+        Lp[e][r][c] =
+            L[((long)(d0 + e) * mp + rb * 16 + r) * mp + c];
       }
       __syncthreads();
     }
-    f64x4 acc = -W[rb];  // -RHS_rb
+    f64x4 acc[DPG];
+#pragma unroll
+    for (int e = 0; e < DPG; ++e) acc[e] = -W[e][rb];  // -RHS_rb
 #pragma unroll
     for (int cb = 0; cb < rb; ++cb) {
 #pragma unroll
       for (int kk = 0; kk < 4; ++kk) {
-        const double a = Lp[li][cb * 16 + kk * 4 + lk];
-        acc = MFMA_F64(a, W[cb][kk], acc);  // b-frag = reg kk (identity)
+#pragma unroll
+        for (int e = 0; e < DPG; ++e) {
+          const double a = Lp[e][li][cb * 16 + kk * 4 + lk];
+          acc[e] = MFMA_F64(a, W[e][cb][kk], acc[e]);
+        }
       }
     }
     // W[rb] = Iv[rb] * (RHS - sum) = Iv[rb] * (-acc)
-    f64x4 sol = {0, 0, 0, 0};
+    f64x4 sol[DPG];
+#pragma unroll
+    for (int e = 0; e < DPG; ++e) sol[e] = f64x4{0, 0, 0, 0};
 #pragma unroll
     for (int kk = 0; kk < 4; ++kk) {
-      const double a = Iv[rb][li][kk * 4 + lk];
-      sol = MFMA_F64(a, -acc[kk], sol);
+#pragma unroll
+      for (int e = 0; e < DPG; ++e) {
+        const double a = Iv[e][rb][li][kk * 4 + lk];
+        sol[e] = MFMA_F64(a, -acc[e][kk], sol[e]);
+      }
     }
-    W[rb] = sol;
+#pragma unroll
+    for (int e = 0; e < DPG; ++e) W[e][rb] = sol[e];
   }
   __syncthreads();
 
-  // fused reduction: stage one 16-row tile at a time, accumulate the
-  // five per-frequency dots.  Lane (segment lk, row li) accumulates
-  // freq q = wv*8 + pass*4 + lk over rows li, 16+li, 32+li, ...
-  double pss[2] = {0, 0}, pcc[2] = {0, 0}, psc[2] = {0, 0},
-         psu[2] = {0, 0}, pcu[2] = {0, 0};
+  // fused reduction, register-resident: a frequency's sin/cos columns
+  // are ADJACENT lanes (li even/odd), so the per-frequency dots come
+  // from __shfl_xor(w, 1) products accumulated in registers across
+  // (rt, v); only the solved u column (wave 7, li == 14) goes through a
+  // tiny LDS stage.  No barriers inside the accumulation loop (the
+  // LDS-staged 16-row variant cost 2*NBT barriers per draw and
+  // dominated the kernel at small NBT).
+  if (wv == 7) {
 #pragma unroll
-  for (int rt = 0; rt < NBT; ++rt) {
+    for (int rt = 0; rt < NBT; ++rt)
 #pragma unroll
-    for (int v = 0; v < 4; ++v) Wred[4 * v + lk][jw + li] = W[rt][v];
-    __syncthreads();
+      for (int v = 0; v < 4; ++v)
 #pragma unroll
-    for (int pass = 0; pass < 2; ++pass) {
-      const int q = wv * 8 + pass * 4 + lk;
-      if (q < FPT_FREQS && f0 + q < F) {
-        const double ws = Wred[li][2 * q];
-        const double wc = Wred[li][2 * q + 1];
-        const double wu = Wred[li][126];
-        pss[pass] = fma(ws, ws, pss[pass]);
-        pcc[pass] = fma(wc, wc, pcc[pass]);
-        psc[pass] = fma(ws, wc, psc[pass]);
-        psu[pass] = fma(ws, wu, psu[pass]);
-        pcu[pass] = fma(wc, wu, pcu[pass]);
+        for (int e = 0; e < DPG; ++e)
+          if (li == 14) Wu[e][rt * 16 + 4 * v + lk] = W[e][rt][v];
+  }
+  __syncthreads();
+
+#pragma unroll
+  for (int e = 0; e < DPG; ++e) {
+    if (e >= ndr) break;
+    double pss = 0, pcc = 0, psc = 0, psu = 0, pcu = 0;
+#pragma unroll
+    for (int rt = 0; rt < NBT; ++rt) {
+#pragma unroll
+      for (int v = 0; v < 4; ++v) {
+        const double w = W[e][rt][v];
+        const double wp = __shfl_xor(w, 1, 64);  // partner column
+        const double wu = Wu[e][rt * 16 + 4 * v + lk];
+        // even lanes own the (sin, cos) pair of their frequency
+        pss = fma(w, w, pss);
+        pcc = fma(wp, wp, pcc);
+        psc = fma(w, wp, psc);
+        psu = fma(w, wu, psu);
+        pcu = fma(wp, wu, pcu);
       }
     }
-    __syncthreads();
-  }
-#pragma unroll
-  for (int pass = 0; pass < 2; ++pass) {
-#pragma unroll
-    for (int off = 8; off > 0; off >>= 1) {
-      pss[pass] += __shfl_down(pss[pass], off, 16);
-      pcc[pass] += __shfl_down(pcc[pass], off, 16);
-      psc[pass] += __shfl_down(psc[pass], off, 16);
-      psu[pass] += __shfl_down(psu[pass], off, 16);
-      pcu[pass] += __shfl_down(pcu[pass], off, 16);
-    }
-    const int q = wv * 8 + pass * 4 + lk;
-    if (li == 0 && q < FPT_FREQS && f0 + q < F) {
+    // sum the 4 row groups (lanes lk = 0..3 share li): +16, +32 lanes
+    pss += __shfl_down(pss, 32, 64); pss += __shfl_down(pss, 16, 64);
+    pcc += __shfl_down(pcc, 32, 64); pcc += __shfl_down(pcc, 16, 64);
+    psc += __shfl_down(psc, 32, 64); psc += __shfl_down(psc, 16, 64);
+    psu += __shfl_down(psu, 32, 64); psu += __shfl_down(psu, 16, 64);
+    pcu += __shfl_down(pcu, 32, 64); pcu += __shfl_down(pcu, 16, 64);
+    const int q = (jw + li) >> 1;  // frequency index within the block
+    if (lk == 0 && (li & 1) == 0 && q < FPT_FREQS && f0 + q < F) {
       const int f = f0 + q;
-      const double M11 = sNs[f] - gsign * pss[pass];
-      const double M22 = sNs[F + f] - gsign * pcc[pass];
-      const double M12 = sNs[2 * F + f] - gsign * psc[pass];
-      const double N1 = sNr[f] - gsign * psu[pass];
-      const double N2 = sNr[F + f] - gsign * pcu[pass];
+      const double M11 = sNs[f] - gsign * pss;
+      const double M22 = sNs[F + f] - gsign * pcc;
+      const double M12 = sNs[2 * F + f] - gsign * psc;
+      const double N1 = sNr[f] - gsign * psu;
+      const double N2 = sNr[F + f] - gsign * pcu;
       const double det = fma(M11, M22, -M12 * M12);
       const double num =
           fma(N1 * N1, M22, fma(-2.0 * N1, N2 * M12, N2 * N2 * M11));
-      fp[(long)d * F + f] += 0.5 * num / det;
+      fp[(long)(d0 + e) * F + f] += 0.5 * num / det;
     }
   }
 }
@@ -661,7 +699,7 @@ void launch_sbgemm(const double* T, const double* toas, const double* ninv,
                    double* out, long plane_stride, long ldo, int ksplit,
                    hipStream_t stream) {
   const int ctiles = (F2 + 63) / 64;
-  hipLaunchKernelGGL(sbgemm_kernel, dim3(ctiles, ksplit), dim3(256), 0,
+  hipLaunchKernelGGL(sbgemm_kernel, dim3(ctiles, ksplit), dim3(512), 0,
                      stream, T, toas, ninv, freqs, ntoa, m, mp, F2, out,
                      plane_stride, ldo);
 }
@@ -683,14 +721,27 @@ void launch_trsm_fp(const double* L, const double* invd, const double* RHS,
                     const double* sNs, const double* sNr, int mp, int F,
                     int D, double gsign, double* fp, hipStream_t stream) {
   const int ftiles = (F + FPT_FREQS - 1) / FPT_FREQS;
-  const dim3 grid(ftiles, D), blk(512);
-  switch (mp >> 4) {
+  const dim3 blk(512);
+  const int nb = mp >> 4;
+  // two draws per workgroup when the solve is small (compressed path)
+  if (nb <= 4) {
+    const dim3 grid(ftiles, (D + 1) / 2);
+    switch (nb) {
+#define TRSM_CASE2(NBT) \
+      case NBT: hipLaunchKernelGGL((trsm_fp_kernel<NBT, 2>), grid, blk, 0, \
+                    stream, L, invd, RHS, sNs, sNr, F, D, gsign, fp); break;
+      TRSM_CASE2(1) TRSM_CASE2(2) TRSM_CASE2(3) TRSM_CASE2(4)
+#undef TRSM_CASE2
+    }
+  } else {
+    const dim3 grid(ftiles, D);
+    switch (nb) {
 #define TRSM_CASE(NBT) \
-    case NBT: hipLaunchKernelGGL(trsm_fp_kernel<NBT>, grid, blk, 0, stream, \
-                                 L, invd, RHS, sNs, sNr, F, D, gsign, fp); break;
-    TRSM_CASE(1) TRSM_CASE(2) TRSM_CASE(3) TRSM_CASE(4)
-    TRSM_CASE(5) TRSM_CASE(6) TRSM_CASE(7) TRSM_CASE(8)
+      case NBT: hipLaunchKernelGGL((trsm_fp_kernel<NBT, 1>), grid, blk, 0, \
+                    stream, L, invd, RHS, sNs, sNr, F, D, gsign, fp); break;
+      TRSM_CASE(5) TRSM_CASE(6) TRSM_CASE(7) TRSM_CASE(8)
 #undef TRSM_CASE
+    }
   }
 }
 
