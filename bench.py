@@ -37,6 +37,21 @@ from fastfp_amd.noise import batch_phiinv, check_batch_homogeneous
 from fastfp_amd.parallel import all_gather_concat, cleanup, init_distributed
 
 
+#: BASELINE.json config presets (see BASELINE.md / docs/DESIGN.md §5)
+PRESETS = {
+    # config 3 (default): 67-psr NMFp, 1e3 freqs, draw batches
+    "nmfp67": {},
+    # config 2: 45-psr 1e4-frequency Fp sweep, one noise draw
+    # (draw compression only pays for D > 1)
+    "fp45": dict(npsr=45, freqs=10000, draws_per_step=1, no_compress=True),
+    # config 4: 67-psr NMFp with ECORR as block-diagonal white noise
+    "ecorr67": dict(ecorr_kernel=True),
+    # config 5: 200-psr SKA-scale, 1e5-frequency sweep (HBM sizing)
+    "ska200": dict(npsr=200, ntoa=10000, freqs=100000, draws_per_step=1,
+                   no_compress=True),
+}
+
+
 def build_problem(args, device, rank):
     psrs = make_synthetic_pta(
         npsr=args.npsr,
@@ -50,8 +65,12 @@ def build_problem(args, device, rank):
     for p in psrs:
         noise[f"{p.name}_red_noise_gamma"] = 13.0 / 3.0
         noise[f"{p.name}_red_noise_log10_A"] = -14.5
+        if args.ecorr_kernel:
+            for b in np.unique(p.backend_flags):
+                noise[f"{p.name}_basis_ecorr_{b}_log10_ecorr"] = -6.5
     pta = initialize_pta(
-        psrs, noise, inc_cp=True, rn_comps=args.rn_comps, gwb_comps=args.gwb_comps
+        psrs, noise, inc_cp=True, rn_comps=args.rn_comps,
+        gwb_comps=args.gwb_comps, ecorr_kernel=args.ecorr_kernel,
     )
     TNTs, Nvecs, Ts = get_mats_nmfp(pta, noise)
 
@@ -110,7 +129,16 @@ def main():
                     help="disable hipGraph capture of the step")
     ap.add_argument("--no-compress", action="store_true",
                     help="disable the Schur draw compression")
+    ap.add_argument("--ecorr-kernel", action="store_true",
+                    help="ECORR as block-diagonal white noise (config 4)")
+    ap.add_argument("--preset", choices=sorted(PRESETS), default=None,
+                    help="BASELINE.json config preset")
     args = ap.parse_args()
+    if args.preset:
+        defaults = {a.dest: a.default for a in ap._actions}
+        for k, v in PRESETS[args.preset].items():
+            if getattr(args, k) == defaults.get(k):
+                setattr(args, k, v)
 
     rank, world, device = init_distributed(
         device=torch.device(args.device) if args.device else None

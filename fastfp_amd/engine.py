@@ -216,25 +216,42 @@ class FpEngine:
             sigma0 = blk.TNT + torch.diag(pf)
             L0 = torch.linalg.cholesky(sigma0)
             RHSe = blk.RHS[:m, :]  # (m, 2F+1)
-            W0 = torch.linalg.solve_triangular(L0, RHSe, upper=False)
-            wu = W0[:, -1]
-            Ws = W0[:, 0:-1:2]
-            Wc = W0[:, 1:-1:2]
-            M0 = torch.stack(
-                [
-                    blk.sNs[0] - (Ws * Ws).sum(0),
-                    blk.sNs[1] - (Wc * Wc).sum(0),
-                    blk.sNs[2] - (Ws * Wc).sum(0),
-                ]
-            ).contiguous()
-            N0 = torch.stack(
-                [blk.sNr[0] - Ws.T @ wu, blk.sNr[1] - Wc.T @ wu]
-            ).contiguous()
-            S0iR = torch.linalg.solve_triangular(
-                L0.transpose(0, 1), W0, upper=True
+            ncols = RHSe.shape[1]
+            mv = len(range(*sl.indices(m)))
+            # column-chunked solves: rocBLAS trsm wants per-call
+            # workspace ~ n, which fails outright at the 2e5-frequency
+            # SKA shape
+            K = torch.empty((mv, ncols), dtype=torch.float64, device=self.device)
+            M0 = torch.empty((3, ncols // 2), dtype=torch.float64, device=self.device)
+            N0 = torch.empty((2, ncols // 2), dtype=torch.float64, device=self.device)
+            CH = 16384  # even: chunks stay aligned to sin/cos pairs
+            # first pass: the u column (last), needed by every chunk
+            wcol = torch.linalg.solve_triangular(
+                L0, RHSe[:, -1:], upper=False
             )
-            K = S0iR[sl, :]  # (mv, 2F+1)
-            mv = K.shape[0]
+            for lo in range(0, ncols, CH):
+                hi = min(lo + CH, ncols)
+                W0c = torch.linalg.solve_triangular(
+                    L0, RHSe[:, lo:hi], upper=False
+                )
+                # frequency-column bookkeeping: chunks are aligned to
+                # even columns (CH is even), so cols lo..hi-1 pair up;
+                # the final chunk also carries the (already solved) u col
+                fs, fe = lo // 2, (hi - (1 if hi == ncols else 0)) // 2
+                Ws = W0c[:, 0 : 2 * (fe - fs) : 2]
+                Wc = W0c[:, 1 : 2 * (fe - fs) : 2]
+                wu = wcol[:, 0]
+                M0[0, fs:fe] = blk.sNs[0, fs:fe] - (Ws * Ws).sum(0)
+                M0[1, fs:fe] = blk.sNs[1, fs:fe] - (Wc * Wc).sum(0)
+                M0[2, fs:fe] = blk.sNs[2, fs:fe] - (Ws * Wc).sum(0)
+                N0[0, fs:fe] = blk.sNr[0, fs:fe] - Ws.T @ wu
+                N0[1, fs:fe] = blk.sNr[1, fs:fe] - Wc.T @ wu
+                S0c = torch.linalg.solve_triangular(
+                    L0.transpose(0, 1), W0c, upper=True
+                )
+                K[:, lo:hi] = S0c[sl, :]
+            M0 = M0.contiguous()
+            N0 = N0.contiguous()
             S0inv = torch.cholesky_inverse(L0)
             G = S0inv[sl, sl].contiguous()
             if self._use_hip:
