@@ -91,6 +91,21 @@ def create_quantization_matrix(
     return U, weights
 
 
+def create_quantization_array(toas, dt: float = day, nmin: int = 2):
+    """Per-epoch quantization weights only — name/signature parity with
+    the reference helper (``/root/reference/examples/run_nmfp.py:38-57``;
+    weights hardcoded to 1.0 there and here)."""
+    _, weights = create_quantization_matrix(toas, dt=dt, nmin=nmin)
+    return weights
+
+
+def ecorr_weights_by_backend(psr, dt: float = day, nmin: int = 2) -> list:
+    """Quantization weights split by receiver backend — parity with
+    ``/root/reference/examples/run_nmfp.py:60-70``."""
+    _, weights = ecorr_basis_by_backend(psr, dt=dt, nmin=nmin)
+    return weights
+
+
 def ecorr_basis_by_backend(psr, dt: float = day, nmin: int = 2) -> tuple:
     """Build the per-backend ECORR quantization basis for one pulsar.
 

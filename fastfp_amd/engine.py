@@ -33,10 +33,26 @@ back silently to eager on GPU is an error.
 
 from __future__ import annotations
 
+import contextlib
 import math
+import os
 
 import numpy as np
 import torch
+
+
+@contextlib.contextmanager
+def _roctx(name: str):
+    """Optional roctx ranges (shown in rocprof timelines) around the
+    engine phases; enable with FASTFP_ROCTX=1."""
+    if os.environ.get("FASTFP_ROCTX") == "1" and torch.cuda.is_available():
+        torch.cuda.nvtx.range_push(name)
+        try:
+            yield
+        finally:
+            torch.cuda.nvtx.range_pop()
+    else:
+        yield
 
 
 def _t64(x, device):
@@ -126,11 +142,12 @@ class FpEngine:
         freqs = _t64(freqs, self.device).reshape(-1)
         self.freqs = freqs
         F = freqs.shape[0]
-        for blk in self.blocks:
-            if self._use_hip:
-                self._precompute_hip(blk, freqs, freq_chunk)
-            else:
-                self._precompute_eager(blk, freqs, freq_chunk)
+        with _roctx("fastfp:freq_precompute"):
+            for blk in self.blocks:
+                if self._use_hip:
+                    self._precompute_hip(blk, freqs, freq_chunk)
+                else:
+                    self._precompute_eager(blk, freqs, freq_chunk)
         return self
 
     def _precompute_eager(self, blk: PulsarBlock, freqs, freq_chunk):

@@ -290,6 +290,18 @@ class RNContainer:
     def get_phi_rn(self, pars):
         return self._powerlaw(pars)
 
+    def get_phi_rn_curn(self, pars):
+        assert not self.inc_tm and not self.gp_ecorr and self.add_curn
+        return self.update_phi(pars)
+
+    def get_phi_ecorr_rn(self, pars):
+        assert not self.inc_tm and self.gp_ecorr and not self.add_curn
+        return self.update_phi(pars)
+
+    def get_phi_ecorr_rn_curn(self, pars):
+        assert not self.inc_tm and self.gp_ecorr and self.add_curn
+        return self.update_phi(pars)
+
     def get_phi_tm_rn(self, pars):
         assert self.inc_tm and not self.gp_ecorr and not self.add_curn
         return self.update_phi(pars)
