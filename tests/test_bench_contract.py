@@ -1,0 +1,37 @@
+"""The driver depends on bench.py's CLI + one-line JSON contract."""
+
+import json
+import os
+import subprocess
+import sys
+
+REPO = os.path.abspath(os.path.join(os.path.dirname(__file__), ".."))
+
+
+def _run(*extra):
+    out = subprocess.run(
+        [sys.executable, "bench.py", "--npsr", "2", "--ntoa", "120",
+         "--ntm", "4", "--rn-comps", "3", "--gwb-comps", "3",
+         "--freqs", "8", "--draws-per-step", "4", "--steps", "1",
+         "--warmup", "0", "--device", "cpu", *extra],
+        cwd=REPO, capture_output=True, text=True, timeout=300,
+    )
+    assert out.returncode == 0, out.stderr[-2000:]
+    line = out.stdout.strip().splitlines()[-1]
+    return json.loads(line)
+
+def test_bench_json_contract():
+    j = _run()
+    for key in ("metric", "value", "unit", "n_gpus", "steps", "warmup",
+                "ms_per_step", "higher_is_better", "scaling",
+                "vs_baseline", "dtype", "data", "config"):
+        assert key in j, key
+    assert j["n_gpus"] == 1 and j["steps"] == 1
+    assert j["dtype"] == "fp64" and j["data"] == "synthetic"
+    assert j["higher_is_better"] is True and j["scaling"] == "weak"
+    assert j["value"] > 0 and j["vs_baseline"] == j["value"] / 6.0
+    assert j["config"]["global_batch"] == 4
+
+def test_bench_presets_parse():
+    j = _run("--preset", "ecorr67")
+    assert j["value"] > 0

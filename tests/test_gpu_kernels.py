@@ -83,11 +83,10 @@ def test_sbgemm_vs_torch(F, ntoa):
     torch.testing.assert_close(RHS[:m, -1], TNr, rtol=1e-12, atol=0.0)
 
 
-def test_chol_batch_vs_torch():
+@pytest.mark.parametrize("m,mp", [(37, 48), (100, 112), (120, 128)])
+def test_chol_batch_vs_torch(m, mp):
     ext = _ext()
-    toas, nvec, r, T, freqs, phiinv = _rand_problem()
-    m = T.shape[1]
-    mp = 48  # padded
+    toas, nvec, r, T, freqs, phiinv = _rand_problem(m=m)
     TN = T / nvec[:, None]
     TNT = (T.T @ TN).contiguous()
     L, invd = ext.chol_batch(TNT, phiinv, mp)
@@ -111,10 +110,11 @@ def test_chol_batch_vs_torch():
         )
 
 
-def test_trsm_fp_accum_vs_eager():
+@pytest.mark.parametrize("m", [37, 120])  # DPG=2 and direct NBT=8 paths
+def test_trsm_fp_accum_vs_eager(m):
     from fastfp_amd import ops
 
-    toas, nvec, r, T, freqs, phiinv = _rand_problem(F=100, D=7)
+    toas, nvec, r, T, freqs, phiinv = _rand_problem(F=100, D=7, m=m)
     m = T.shape[1]
     F = freqs.shape[0]
     D = phiinv.shape[0]
