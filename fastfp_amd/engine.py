@@ -126,11 +126,16 @@ class FpEngine:
         ]
         self.freqs = None
         self._use_hip = False
+        self._side_stream = None
         if self.device.type == "cuda" and not force_eager:
             from fastfp_amd import ops
 
             ops.require_hip()  # fail loudly if the extension is missing
             self._use_hip = True
+            # second stream: per-pulsar chol->trsm chains are independent,
+            # so alternating pulsars across two streams overlaps each
+            # pulsar's Cholesky with the previous pulsar's solve
+            self._side_stream = torch.cuda.Stream(device=self.device)
 
     # ------------------------------------------------------------------
     # frequency precompute
@@ -337,45 +342,68 @@ class FpEngine:
         if fp is None:
             fp = torch.zeros((D, F), dtype=torch.float64, device=self.device)
 
+        fp_side = None
+        main_stream = None
+        if self._use_hip:
+            main_stream = torch.cuda.current_stream(self.device)
+            fp_side = torch.zeros_like(fp)
+            ev = torch.cuda.Event()
+            ev.record(main_stream)
+            self._side_stream.wait_event(ev)
+
         for lo in range(0, D, draw_chunk):
             hi = min(lo + draw_chunk, D)
             for i, blk in enumerate(self.blocks):
-                if phiinvs is not None:
-                    pinv = _t64(phiinvs[i], self.device)
-                    pinv = pinv[None, :] if pinv.dim() == 1 else pinv[lo:hi]
-                    sigma = None
-                else:
-                    sg = _t64(sigmas[i], self.device)
-                    sigma = sg[None, :, :] if sg.dim() == 2 else sg[lo:hi]
-                    # get_mats_fp contract: sigma = TNT + diag(phi^-1)
-                    pinv = (
-                        torch.diagonal(sigma, dim1=-2, dim2=-1)
-                        - torch.diagonal(blk.TNT)[None, :]
-                    )
-                if blk.comp is not None:
-                    # Schur-compressed: C_d = diag(1/Delta_d) + G
-                    c = blk.comp
-                    phi_var = (
-                        1.0 / (pinv[:, c["var"]] - c["delta0"][None, :])
-                    ).contiguous()
-                    if self._use_hip:
-                        from fastfp_amd import ops
-
-                        ops.chol_trsm_fp_accum(
-                            c["G"], phi_var, c["K"], c["M0"], c["N0"],
-                            fp[lo:hi], gsign=-1.0,
-                        )
+                side = self._use_hip and (i & 1) == 1
+                stream_ctx = (
+                    torch.cuda.stream(self._side_stream)
+                    if side
+                    else contextlib.nullcontext()
+                )
+                fp_tgt = (fp_side if side else fp)[lo:hi]
+                with stream_ctx:
+                    if phiinvs is not None:
+                        pinv = _t64(phiinvs[i], self.device)
+                        pinv = pinv[None, :] if pinv.dim() == 1 else pinv[lo:hi]
+                        sigma = None
                     else:
-                        sigc = c["G"][None, :, :] + torch.diag_embed(phi_var)
-                        self._accum_eager_mats(
-                            sigc, c["K"], c["M0"], c["N0"], fp[lo:hi], -1.0
+                        sg = _t64(sigmas[i], self.device)
+                        sigma = sg[None, :, :] if sg.dim() == 2 else sg[lo:hi]
+                        # get_mats_fp contract: sigma = TNT + diag(phi^-1)
+                        pinv = (
+                            torch.diagonal(sigma, dim1=-2, dim2=-1)
+                            - torch.diagonal(blk.TNT)[None, :]
                         )
-                elif self._use_hip:
-                    self._accum_hip(blk, pinv.contiguous(), fp[lo:hi])
-                else:
-                    if sigma is None:
-                        sigma = blk.TNT[None, :, :] + torch.diag_embed(pinv)
-                    self._accum_eager(blk, sigma, fp[lo:hi])
+                    if blk.comp is not None:
+                        # Schur-compressed: C_d = diag(1/Delta_d) + G
+                        c = blk.comp
+                        phi_var = (
+                            1.0 / (pinv[:, c["var"]] - c["delta0"][None, :])
+                        ).contiguous()
+                        if self._use_hip:
+                            from fastfp_amd import ops
+
+                            ops.chol_trsm_fp_accum(
+                                c["G"], phi_var, c["K"], c["M0"], c["N0"],
+                                fp_tgt, gsign=-1.0,
+                            )
+                        else:
+                            sigc = c["G"][None, :, :] + torch.diag_embed(phi_var)
+                            self._accum_eager_mats(
+                                sigc, c["K"], c["M0"], c["N0"], fp_tgt, -1.0
+                            )
+                    elif self._use_hip:
+                        self._accum_hip(blk, pinv.contiguous(), fp_tgt)
+                    else:
+                        if sigma is None:
+                            sigma = blk.TNT[None, :, :] + torch.diag_embed(pinv)
+                        self._accum_eager(blk, sigma, fp_tgt)
+
+        if self._use_hip:
+            ev2 = torch.cuda.Event()
+            ev2.record(self._side_stream)
+            main_stream.wait_event(ev2)
+            fp += fp_side
 
         return fp[0] if not batched else fp
 
