@@ -48,7 +48,7 @@ def require_hip() -> None:
         )
 
 
-MAX_MP = 128  # kernel LDS-resident limit on the padded basis size
+MAX_MP = 128  # LDS-resident limit of the chol/trsm SOLVE kernels
 
 
 def pad16(m: int) -> int:
@@ -56,12 +56,18 @@ def pad16(m: int) -> int:
 
 
 def check_m(m: int) -> int:
+    """Solve-dimension check: the batched Cholesky/TRSM kernels keep the
+    matrix (LDS) / RHS strip (registers) on-chip, capped at 128.  The
+    Schur draw compression reduces the per-draw dimension to the
+    variable bins, so large-m models (GP-ECORR) run through NMFp with
+    compression; only the DIRECT per-draw path is capped."""
     mp = pad16(m)
     if mp > MAX_MP:
         raise NotImplementedError(
-            f"GPU kernels currently support basis size m <= {MAX_MP} per "
-            f"pulsar (got m={m}); reduce the ECORR epoch count or use the "
-            "CPU engine for this model"
+            f"the direct per-draw solve supports m <= {MAX_MP} (got m={m}); "
+            "use the Schur-compressed path (NMFp.sweep compress=True, on by "
+            "default) whose per-draw dimension is the variable-bin count, "
+            "or the CPU engine"
         )
     return mp
 
@@ -81,7 +87,7 @@ def freq_precompute(toas, Nvec, r, T, TNr, freqs, freq_chunk: int = 8192):
     ext = _try_load()
     F = int(freqs.shape[0])
     ntoa, m = T.shape
-    mp = check_m(m)
+    mp = pad16(m)  # sbgemm is M-tiled: no 128 cap on the basis size
     device = T.device
     Ninv = (1.0 / Nvec).contiguous()
     Nr = (r / Nvec).contiguous()
@@ -117,7 +123,7 @@ def freq_precompute_block(toas, block_noise, Nr, V, TNr, freqs):
     ext = _try_load()
     F = int(freqs.shape[0])
     ntoa, m = V.shape
-    mp = check_m(m)
+    mp = pad16(m)  # sbgemm is M-tiled: no 128 cap on the basis size
     device = V.device
     freqs = freqs.contiguous()
 

@@ -120,6 +120,8 @@ extern "C" __global__ __launch_bounds__(512) void sbgemm_kernel(
   // 8 accumulator tiles per wave (64+ AGPR) which capped occupancy at
   // 3 waves/SIMD; splitting the M dim halves the accumulator and lets
   // more waves cover the sincos + LDS latency.
+  // grid.z tiles the M dimension in blocks of 128 so the basis size is
+  // unbounded (GP-ECORR models run to many hundreds of columns).
   __shared__ double lT[16][FASTFP_MAXMP + 1];
   __shared__ double lS[64][17];
 
@@ -127,10 +129,12 @@ extern "C" __global__ __launch_bounds__(512) void sbgemm_kernel(
   const int lane = tid & 63;
   const int wv = tid >> 6;
   const int c0 = blockIdx.x * 64;       // first output column of this tile
+  const int mbase = blockIdx.z * FASTFP_MAXMP;  // M-tile origin
+  const int mp_loc = min(FASTFP_MAXMP, mp - mbase);
   const int jw = (wv & 3) * 16;         // column strip
   const int rh = wv >> 2;               // row half (0: rows 0-63, 1: 64-127)
   const int rowbase = rh * 64;
-  const int nrt_tot = mp >> 4;
+  const int nrt_tot = mp_loc >> 4;
   // row tiles this wave owns (half 1 may be empty for small mp)
   const int rt_lo = min(rh * 4, nrt_tot);
   const int rt_hi = min(rt_lo + 4, nrt_tot);
@@ -147,11 +151,12 @@ extern "C" __global__ __launch_bounds__(512) void sbgemm_kernel(
   for (int q = 0; q < 4; ++q) acc[q] = f64x4{0, 0, 0, 0};
 
   for (int k0 = kbeg; k0 < kend; k0 += 16) {
-    // stage T panel rows k0..k0+15 (zero-padded)
-    for (int idx = tid; idx < 16 * mp; idx += 512) {
-      const int k = idx / mp, j = idx % mp;
+    // stage T panel rows k0..k0+15, cols mbase.. (zero-padded)
+    for (int idx = tid; idx < 16 * mp_loc; idx += 512) {
+      const int k = idx / mp_loc, j = idx % mp_loc;
       const int gk = k0 + k;
-      lT[k][j] = (gk < kend && j < m) ? T[(long)gk * m + j] : 0.0;
+      const int gj = mbase + j;
+      lT[k][j] = (gk < kend && gj < m) ? T[(long)gk * m + gj] : 0.0;
     }
     // stage trig panel: 32 freq-pairs x 16 toas, one sincos each
     for (int idx = tid; idx < 32 * 16; idx += 512) {
@@ -190,7 +195,7 @@ extern "C" __global__ __launch_bounds__(512) void sbgemm_kernel(
     if (rt_lo + q >= rt_hi) break;
 #pragma unroll
     for (int v = 0; v < 4; ++v) {
-      const int row = rowbase + q * 16 + 4 * v + (lane >> 4);
+      const int row = mbase + rowbase + q * 16 + 4 * v + (lane >> 4);
       const int col = c0 + jw + (lane & 15);
       if (col < F2) outp[(long)row * ldo + col] = acc[q][v];
     }
@@ -699,8 +704,9 @@ void launch_sbgemm(const double* T, const double* toas, const double* ninv,
                    double* out, long plane_stride, long ldo, int ksplit,
                    hipStream_t stream) {
   const int ctiles = (F2 + 63) / 64;
-  hipLaunchKernelGGL(sbgemm_kernel, dim3(ctiles, ksplit), dim3(512), 0,
-                     stream, T, toas, ninv, freqs, ntoa, m, mp, F2, out,
+  const int mtiles = (mp + FASTFP_MAXMP - 1) / FASTFP_MAXMP;
+  hipLaunchKernelGGL(sbgemm_kernel, dim3(ctiles, ksplit, mtiles), dim3(512),
+                     0, stream, T, toas, ninv, freqs, ntoa, m, mp, F2, out,
                      plane_stride, ldo);
 }
 

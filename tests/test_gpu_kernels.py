@@ -60,11 +60,12 @@ def test_sigdots_vs_torch():
     torch.testing.assert_close(sNr[1], C @ nr, rtol=1e-10, atol=1e-8)
 
 
-@pytest.mark.parametrize("F,ntoa", [(29, 333), (70, 150), (200, 2000)])
-def test_sbgemm_vs_torch(F, ntoa):
+@pytest.mark.parametrize("F,ntoa,m", [(29, 333, 37), (70, 150, 37),
+                                      (200, 2000, 37), (40, 500, 300)])
+def test_sbgemm_vs_torch(F, ntoa, m):
     from fastfp_amd import ops
 
-    toas, nvec, r, T, freqs, _ = _rand_problem(F=F, ntoa=ntoa)
+    toas, nvec, r, T, freqs, _ = _rand_problem(F=F, ntoa=ntoa, m=m)
     ninv = 1.0 / nvec
     TNr = (T / nvec[:, None]).T @ r
     RHS, sNs, sNr = ops.freq_precompute(toas, nvec, r, T, TNr, freqs)
@@ -247,3 +248,37 @@ def test_engine_blocknoise_gpu_matches_cpu():
     cpu = fp_obj.sweep(freqs, Nvecs, Ts, sigmas, device="cpu")
     gpu = fp_obj.sweep(freqs, Nvecs, Ts, sigmas, device=DEV)
     np.testing.assert_allclose(gpu, cpu, rtol=1e-8)
+
+
+def test_nmfp_large_m_ecorr_gpu_compressed():
+    """GP-ECORR model with basis size m > 128: precompute is M-tiled,
+    the per-draw solve runs compressed at the variable-bin dimension."""
+    from fastfp_amd import get_mats_nmfp, initialize_pta, make_synthetic_pta
+    from fastfp_amd.nmfp import NMFp
+
+    psrs = make_synthetic_pta(npsr=2, ntoa=700, ntm=5, seed=21)
+    noise = {}
+    for p in psrs:
+        noise[f"{p.name}_red_noise_gamma"] = 4.0
+        noise[f"{p.name}_red_noise_log10_A"] = -14.5
+        for b in np.unique(p.backend_flags):
+            noise[f"{p.name}_basis_ecorr_{b}_log10_ecorr"] = -6.5
+            noise[f"{p.name}_{b}_efac"] = 1.0
+    pta = initialize_pta(psrs, noise, inc_cp=False, rn_comps=6,
+                         simple_wn=True, inc_ecorr=True)
+    TNTs, Nvecs, Ts = get_mats_nmfp(pta, noise)
+    assert max(T.shape[1] for T in Ts) > 128, "test needs m > 128"
+    D = 3
+    rng = np.random.default_rng(5)
+    samples = {
+        n: (rng.uniform(2, 6, D) if n.endswith("gamma")
+            else rng.uniform(-16, -14, D))
+        for n in pta.params
+    }
+    nm = NMFp(psrs, pta.rn_containers)
+    freqs = np.linspace(4e-9, 5e-8, 9)
+    cpu = nm.sweep(freqs, samples, Nvecs, Ts, device="cpu")
+    for c in pta.rn_containers:
+        c.to(DEV)
+    gpu = nm.sweep(freqs, samples, Nvecs, Ts, device=DEV)
+    np.testing.assert_allclose(gpu, cpu, rtol=1e-7)
