@@ -156,7 +156,14 @@ def main(
 
     t0 = time.perf_counter()
     parts = []
-    for lo in range(0, len(my_idx), batch_size):
+    try:
+        from tqdm import tqdm
+
+        batches = tqdm(range(0, len(my_idx), batch_size),
+                       disable=rank != 0, desc="draw batches")
+    except ImportError:
+        batches = range(0, len(my_idx), batch_size)
+    for lo in batches:
         sel = my_idx[lo : lo + batch_size]
         ck = os.path.join(batch_dir, f"r{rank}_b{lo}.npy")
         if resume and os.path.exists(ck):
