@@ -453,9 +453,14 @@ __global__ __launch_bounds__(512, 4) void trsm_fp_kernel(
       }
       __syncthreads();
     }
-    f64x4 acc[DPG];
+    // two accumulator chains per draw (even/odd cb) deepen the MFMA
+    // pipeline; folded together before the diagonal solve
+    f64x4 acc[DPG], acc2[DPG];
 #pragma unroll
-    for (int e = 0; e < DPG; ++e) acc[e] = -W[e][rb];  // -RHS_rb
+    for (int e = 0; e < DPG; ++e) {
+      acc[e] = -W[e][rb];  // -RHS_rb
+      acc2[e] = f64x4{0, 0, 0, 0};
+    }
 #pragma unroll
     for (int cb = 0; cb < rb; ++cb) {
 #pragma unroll
@@ -463,10 +468,15 @@ __global__ __launch_bounds__(512, 4) void trsm_fp_kernel(
 #pragma unroll
         for (int e = 0; e < DPG; ++e) {
           const double a = Lp[e][li][cb * 16 + kk * 4 + lk];
-          acc[e] = MFMA_F64(a, W[e][cb][kk], acc[e]);
+          if (cb & 1)
+            acc2[e] = MFMA_F64(a, W[e][cb][kk], acc2[e]);
+          else
+            acc[e] = MFMA_F64(a, W[e][cb][kk], acc[e]);
         }
       }
     }
+#pragma unroll
+    for (int e = 0; e < DPG; ++e) acc[e] += acc2[e];
     // W[rb] = Iv[rb] * (RHS - sum) = Iv[rb] * (-acc)
     f64x4 sol[DPG];
 #pragma unroll
