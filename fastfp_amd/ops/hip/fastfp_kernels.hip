@@ -38,6 +38,7 @@
 
 #include <hip/hip_runtime.h>
 #include <math.h>
+#include <stdlib.h>
 
 #define FASTFP_MAXMP 128
 #define NB 16  // Cholesky/TRSM block size (one MFMA tile)
@@ -215,7 +216,7 @@ extern "C" __global__ __launch_bounds__(512) void sbgemm_kernel(
 #define A_(i, j) Ash[(i) * (NBT * 16 + 1) + (j)]
 
 template <int NBT>
-__global__ __launch_bounds__(512) void chol_batch_kernel(
+__global__ __launch_bounds__(512, 4) void chol_batch_kernel(
     const double* __restrict__ TNT /*(m,m)*/,
     const double* __restrict__ phiinv /*(D,m)*/, int m, int D,
     double* __restrict__ L /*(D,mp,mp)*/,
@@ -226,7 +227,7 @@ __global__ __launch_bounds__(512) void chol_batch_kernel(
   // profile showed 84% of wave cycles parked, profiles/).
   constexpr int mp = NBT * 16;
   __shared__ double Ash[mp * (mp + 1)];
-  __shared__ double inv16[16][17];
+  __shared__ double inv16[2][16][17];
 
   const int d = blockIdx.x;
   if (d >= D) return;
@@ -235,8 +236,10 @@ __global__ __launch_bounds__(512) void chol_batch_kernel(
   const int wv = tid >> 6;
   constexpr int nb = NBT;
 
+  const int nthr = blockDim.x;
+  const int nwv = nthr >> 6;
   // assemble Sigma in LDS
-  for (int idx = tid; idx < mp * mp; idx += 512) {
+  for (int idx = tid; idx < mp * mp; idx += nthr) {
     const int i = idx / mp, j = idx % mp;
     double v = (i < m && j < m) ? TNT[(long)i * m + j] : 0.0;
     if (i == j) v += (i < m) ? phiinv[(long)d * m + i] : 1.0;
@@ -244,104 +247,126 @@ __global__ __launch_bounds__(512) void chol_batch_kernel(
   }
   __syncthreads();
 
+  // Look-ahead right-looking Cholesky: the serial 16x16 diagonal
+  // factor of column kb+1 (wave 0, register/shfl) runs CONCURRENTLY
+  // with the MFMA trailing updates of columns kb+2.. (waves 1..7) --
+  // the PMC profile showed 84-88% of wave cycles parked waiting on the
+  // serial diagonal.  Two inverted-diagonal buffers ping-pong by kb
+  // parity.
+#define CHOL_DIAG(KB, DST)                                                   \
+  {                                                                          \
+    const int dk0 = (KB)*NB;                                                 \
+    const int i = lane & 15;                                                 \
+    double row[16];                                                          \
+    _Pragma("unroll") for (int c = 0; c < 16; ++c) row[c] =                  \
+        A_(dk0 + i, dk0 + c);                                                \
+    _Pragma("unroll") for (int t = 0; t < 16; ++t) {                         \
+      const double dv = sqrt(__shfl(row[t], t, 16));                         \
+      const double rdv = 1.0 / dv;                                           \
+      if (i > t) row[t] *= rdv;                                              \
+      else if (i == t) row[t] = dv;                                          \
+      _Pragma("unroll") for (int j = t + 1; j < 16; ++j) {                   \
+        const double ljt = __shfl(row[t], j, 16);                            \
+        if (i >= j) row[j] = fma(-row[t], ljt, row[j]);                      \
+      }                                                                      \
+    }                                                                        \
+    _Pragma("unroll") for (int c = 0; c < 16; ++c) if (c <= i)               \
+        A_(dk0 + i, dk0 + c) = row[c];                                       \
+    const int c = i;                                                         \
+    double diag[16], x[16];                                                  \
+    _Pragma("unroll") for (int r = 0; r < 16; ++r) diag[r] =                 \
+        1.0 / __shfl(row[r], r, 16);                                         \
+    _Pragma("unroll") for (int r = 0; r < 16; ++r) {                         \
+      double acc2 = 0.0;                                                     \
+      _Pragma("unroll") for (int t = 0; t < 16; ++t) {                       \
+        const double lrt = __shfl(row[t], r, 16);                            \
+        if (t >= c && t < r) acc2 = fma(lrt, x[t], acc2);                    \
+      }                                                                      \
+      x[r] = (r < c) ? 0.0 : (r == c) ? diag[r] : -acc2 * diag[r];           \
+    }                                                                        \
+    _Pragma("unroll") for (int r = 0; r < 16; ++r) (DST)[r][c] = x[r];       \
+    if (lane < 16)                                                           \
+      _Pragma("unroll") for (int r = 0; r < 16; ++r)                         \
+          invd[((long)d * nb + (KB)) * 256 + r * 16 + c] = x[r];             \
+  }
+
+  // prologue: factor + invert diagonal block 0
+  if (wv == 0) CHOL_DIAG(0, inv16[0]);
+  __syncthreads();
+
   for (int kb = 0; kb < nb; ++kb) {
     const int k0 = kb * NB;
 
-    // 1) register-resident Cholesky of the 16x16 diagonal block.
-    //    Wave 0: lane segment s (16 lanes) redundantly owns rows i=l&15;
-    //    all cross-row traffic is __shfl (width 16), no LDS in the chain.
-    if (wv == 0) {
-      const int i = lane & 15;
-      double row[16];
-#pragma unroll
-      for (int c = 0; c < 16; ++c) row[c] = A_(k0 + i, k0 + c);
-#pragma unroll
-      for (int t = 0; t < 16; ++t) {
-        const double dv = sqrt(__shfl(row[t], t, 16));
-        const double rdv = 1.0 / dv;  // one software f64 div per step
-        if (i > t) row[t] *= rdv;
-        else if (i == t) row[t] = dv;
-#pragma unroll
-        for (int j = t + 1; j < 16; ++j) {
-          const double ljt = __shfl(row[t], j, 16);
-          if (i >= j) row[j] = fma(-row[t], ljt, row[j]);
-        }
-      }
-#pragma unroll
-      for (int c = 0; c < 16; ++c)
-        if (c <= i) A_(k0 + i, k0 + c) = row[c];
-      // 2) invert the block in registers: lane holds column c = i of
-      //    X = L_kk^-1 (forward substitution; all indices compile-time)
-      const int c = i;
-      double diag[16], x[16];
-#pragma unroll
-      for (int r = 0; r < 16; ++r) diag[r] = 1.0 / __shfl(row[r], r, 16);
-#pragma unroll
-      for (int r = 0; r < 16; ++r) {
-        double acc2 = 0.0;
-#pragma unroll
-        for (int t = 0; t < 16; ++t) {
-          const double lrt = __shfl(row[t], r, 16);
-          if (t >= c && t < r) acc2 = fma(lrt, x[t], acc2);
-        }
-        x[r] = (r < c) ? 0.0 : (r == c) ? diag[r] : -acc2 * diag[r];
-      }
-#pragma unroll
-      for (int r = 0; r < 16; ++r) inv16[r][c] = x[r];
-    }
-    __syncthreads();
-
-    // export invdiag (coalesced)
-    for (int idx = tid; idx < 256; idx += 512) {
-      const int r = idx / 16, c = idx % 16;
-      invd[((long)d * nb + kb) * 256 + idx] = inv16[r][c];
-    }
-
-    // 3) panel TRSM: row tiles below the diagonal, P <- P * inv(L_kk)^T
-    for (int rt = kb + 1 + wv; rt < nb; rt += 8) {
+    // panel TRSM: row tiles below the diagonal, P <- P * inv(L_kk)^T
+    for (int rt = kb + 1 + wv; rt < nb; rt += nwv) {
       const int r0 = rt * NB;
       f64x4 pacc = {0, 0, 0, 0};
 #pragma unroll
       for (int kk = 0; kk < 4; ++kk) {
         const double a = A_(r0 + (lane & 15), k0 + kk * 4 + (lane >> 4));
-        const double b = inv16[lane & 15][kk * 4 + (lane >> 4)];  // inv^T[k][j]
+        const double b = inv16[kb & 1][lane & 15][kk * 4 + (lane >> 4)];
         pacc = MFMA_F64(a, b, pacc);
       }
-      // in-wave: all reads above complete before these writes
 #pragma unroll
       for (int v = 0; v < 4; ++v)
         A_(r0 + 4 * v + (lane >> 4), k0 + (lane & 15)) = pacc[v];
     }
     __syncthreads();
+    if (kb + 1 >= nb) break;
 
-    // 4) trailing update: tiles (ib, jb), kb < jb <= ib < nb
-    const int t = nb - kb - 1;
-    const int ntile = t * (t + 1) / 2;
-    for (int q = wv; q < ntile; q += 8) {
-      // triangular index -> (ib, jb), row-major over the lower wedge
-      int ib = kb + 1, rem = q;
-      while (rem > ib - kb - 1) { rem -= (ib - kb); ++ib; }
-      const int jb = kb + 1 + rem;
-      const int i0 = ib * NB, j0 = jb * NB;
+    // trailing phase A1: ONLY column kb+1 tiles, so its diagonal can
+    // factor next (all 8 waves)
+    const int j1 = (kb + 1) * NB;
+    for (int ib = kb + 1 + wv; ib < nb; ib += nwv) {
+      const int i0 = ib * NB;
       f64x4 uacc;
 #pragma unroll
       for (int v = 0; v < 4; ++v)
-        uacc[v] = A_(i0 + 4 * v + (lane >> 4), j0 + (lane & 15));
+        uacc[v] = A_(i0 + 4 * v + (lane >> 4), j1 + (lane & 15));
 #pragma unroll
       for (int kk = 0; kk < 4; ++kk) {
         const double a = -A_(i0 + (lane & 15), k0 + kk * 4 + (lane >> 4));
-        const double b = A_(j0 + (lane & 15), k0 + kk * 4 + (lane >> 4));  // P[jb]^T
+        const double b = A_(j1 + (lane & 15), k0 + kk * 4 + (lane >> 4));
         uacc = MFMA_F64(a, b, uacc);
       }
 #pragma unroll
       for (int v = 0; v < 4; ++v)
-        A_(i0 + 4 * v + (lane >> 4), j0 + (lane & 15)) = uacc[v];
+        A_(i0 + 4 * v + (lane >> 4), j1 + (lane & 15)) = uacc[v];
+    }
+    __syncthreads();
+
+    // phase A2: wave 0 factors diagonal kb+1 while waves 1..7 do the
+    // remaining trailing tiles (jb >= kb+2)
+    if (wv == 0) {
+      CHOL_DIAG(kb + 1, inv16[(kb + 1) & 1]);
+    } else {
+      const int t = nb - kb - 2;
+      const int ntile = t * (t + 1) / 2;
+      for (int q = wv - 1; q < ntile; q += nwv - 1) {
+        int ib = kb + 2, rem = q;
+        while (rem > ib - kb - 2) { rem -= (ib - kb - 1); ++ib; }
+        const int jb = kb + 2 + rem;
+        const int i0 = ib * NB, j0 = jb * NB;
+        f64x4 uacc;
+#pragma unroll
+        for (int v = 0; v < 4; ++v)
+          uacc[v] = A_(i0 + 4 * v + (lane >> 4), j0 + (lane & 15));
+#pragma unroll
+        for (int kk = 0; kk < 4; ++kk) {
+          const double a = -A_(i0 + (lane & 15), k0 + kk * 4 + (lane >> 4));
+          const double b = A_(j0 + (lane & 15), k0 + kk * 4 + (lane >> 4));
+          uacc = MFMA_F64(a, b, uacc);
+        }
+#pragma unroll
+        for (int v = 0; v < 4; ++v)
+          A_(i0 + 4 * v + (lane >> 4), j0 + (lane & 15)) = uacc[v];
+      }
     }
     __syncthreads();
   }
 
   // write back L (full rows; upper-triangle junk is never read)
-  for (int idx = tid; idx < mp * mp; idx += 512) {
+  for (int idx = tid; idx < mp * mp; idx += nthr) {
     const int i = idx / mp, j = idx % mp;
     L[((long)d * mp + i) * mp + j] = A_(i, j);
   }
@@ -722,7 +747,9 @@ void launch_sbgemm(const double* T, const double* toas, const double* ninv,
 
 void launch_chol_batch(const double* TNT, const double* phiinv, int m, int mp,
                        int D, double* L, double* invd, hipStream_t stream) {
-  const dim3 grid(D), blk(512);
+  // small matrices: 256 threads -> 4 workgroups/CU despite the 115-VGPR
+  // diagonal-factor pressure; large: 512 threads for MFMA coverage
+  const dim3 grid(D), blk(mp <= 64 ? 256 : 512);
   switch (mp >> 4) {
 #define CHOL_CASE(NBT) \
     case NBT: hipLaunchKernelGGL(chol_batch_kernel<NBT>, grid, blk, 0, \
@@ -739,8 +766,11 @@ void launch_trsm_fp(const double* L, const double* invd, const double* RHS,
   const int ftiles = (F + FPT_FREQS - 1) / FPT_FREQS;
   const dim3 blk(512);
   const int nb = mp >> 4;
-  // two draws per workgroup when the solve is small (compressed path)
-  if (nb <= 4) {
+  // two draws per workgroup when the solve is small (compressed path);
+  // FASTFP_TRSM_DPG=1 selects the single-draw variant (A/B tuning)
+  static const char* dpg_env = getenv("FASTFP_TRSM_DPG");
+  static const int dpg_want = dpg_env ? atoi(dpg_env) : 1;  // measured: DPG1 (8 waves/SIMD, 61 VGPR) beats DPG2 (4 waves/SIMD) by ~5%
+  if (nb <= 4 && dpg_want >= 2) {
     const dim3 grid(ftiles, (D + 1) / 2);
     switch (nb) {
 #define TRSM_CASE2(NBT) \
@@ -755,6 +785,7 @@ void launch_trsm_fp(const double* L, const double* invd, const double* RHS,
 #define TRSM_CASE(NBT) \
       case NBT: hipLaunchKernelGGL((trsm_fp_kernel<NBT, 1>), grid, blk, 0, \
                     stream, L, invd, RHS, sNs, sNr, F, D, gsign, fp); break;
+      TRSM_CASE(1) TRSM_CASE(2) TRSM_CASE(3) TRSM_CASE(4)
       TRSM_CASE(5) TRSM_CASE(6) TRSM_CASE(7) TRSM_CASE(8)
 #undef TRSM_CASE
     }
