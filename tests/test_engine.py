@@ -179,3 +179,32 @@ def test_draw_compression_matches_direct():
     direct = nm.sweep(freqs, samples, Nvecs, Ts, device="cpu", compress=False)
     comp = nm.sweep(freqs, samples, Nvecs, Ts, device="cpu", compress=True)
     np.testing.assert_allclose(comp, direct, rtol=1e-7)
+
+
+@pytest.mark.parametrize("seed,inc_cp,inc_ecorr,simple_wn,rn", [
+    (11, True, False, True, 3),
+    (12, False, True, True, 4),
+    (13, True, True, False, 5),
+    (14, False, False, False, 6),
+])
+def test_engine_fuzz_configs(seed, inc_cp, inc_ecorr, simple_wn, rn):
+    """Random model configurations: engine sweep == get_xCy parity path."""
+    psrs = make_synthetic_pta(npsr=2, ntoa=80, ntm=3, seed=seed)
+    noise = {"gw_gamma": 13.0 / 3.0, "gw_log10_A": float(np.log10(2e-15))}
+    rng = np.random.default_rng(seed)
+    for p in psrs:
+        noise[f"{p.name}_red_noise_gamma"] = rng.uniform(2, 6)
+        noise[f"{p.name}_red_noise_log10_A"] = rng.uniform(-16, -14)
+        for b in np.unique(p.backend_flags):
+            noise[f"{p.name}_basis_ecorr_{b}_log10_ecorr"] = -6.5
+            noise[f"{p.name}_{b}_efac"] = rng.uniform(0.9, 1.3)
+            noise[f"{p.name}_{b}_log10_t2equad"] = -6.8
+    pta = initialize_pta(psrs, noise, inc_cp=inc_cp, rn_comps=rn,
+                         gwb_comps=min(rn, 3), simple_wn=simple_wn,
+                         inc_ecorr=inc_ecorr)
+    Nvecs, Ts, sigmas = get_mats_fp(pta, noise)
+    freqs = np.linspace(4e-9, 5e-8, 4)
+    fp_obj = FastFp(psrs)
+    want = np.array([fp_obj.calculate_Fp(f, Nvecs, Ts, sigmas) for f in freqs])
+    got = fp_obj.sweep(freqs, Nvecs, Ts, sigmas, device="cpu")
+    np.testing.assert_allclose(got, want, rtol=1e-6)
