@@ -110,3 +110,7 @@ class NMFp:
             engine.disable_draw_compression()
         fp = engine.sweep(phiinvs=phiinvs, draw_chunk=draw_chunk)
         return fp.cpu().numpy()
+
+
+#: reference-compatible class name
+NMFP = NMFp

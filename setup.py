@@ -8,7 +8,7 @@ repo snapshot to GPU boxes (it is git-ignored but NOT gpurun-ignored).
 
 import os
 
-from setuptools import setup
+from setuptools import find_packages, setup
 
 os.environ.setdefault("PYTORCH_ROCM_ARCH", "gfx950")
 
@@ -31,7 +31,7 @@ ext = CUDAExtension(
 setup(
     name="fastfp_amd",
     version="0.1.0",
-    packages=["fastfp_amd"],
+    packages=find_packages(include=["fastfp_amd*"]),
     ext_modules=[ext],
     cmdclass={"build_ext": BuildExtension.with_options(use_ninja=True)},
 )
