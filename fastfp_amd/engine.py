@@ -289,7 +289,35 @@ class FpEngine:
                 G=G, K=K.contiguous(), M0=M0, N0=N0, var=sl, mv=mv,
                 delta0=delta0,
             )
+        self._stack_compression()
         return self
+
+    def _stack_compression(self):
+        """When every pulsar compresses to the same variable dimension,
+        stack the per-pulsar G/K/M0/N0 so the whole sweep is ONE chol
+        launch + ONE trsm launch per draw chunk (grid.z = pulsar) — no
+        per-pulsar launch gaps or tail quantization.  Per-pulsar Fp
+        planes are summed deterministically by torch."""
+        self._comp_stack = None
+        if not self._use_hip:
+            return
+        comps = [blk.comp for blk in self.blocks]
+        if any(c is None for c in comps):
+            return
+        mvs = {c["G"].shape[0] for c in comps}
+        kshapes = {tuple(c["K"].shape) for c in comps}
+        if len(mvs) != 1 or len(kshapes) != 1:
+            return
+        vars_ = [c["var"] for c in comps]
+        self._comp_stack = dict(
+            G=torch.stack([c["G"] for c in comps]).contiguous(),
+            K=torch.stack([c["K"] for c in comps]).contiguous(),
+            M0=torch.stack([c["M0"] for c in comps]).contiguous(),
+            N0=torch.stack([c["N0"] for c in comps]).contiguous(),
+            delta0=torch.stack([c["delta0"] for c in comps]).contiguous(),
+            vars=vars_,
+            mv=comps[0]["G"].shape[0],
+        )
 
     def compression_margin(self, phiinvs) -> float:
         """min over pulsars/draws/bins of phiinv_d / delta_0.  The
@@ -341,6 +369,10 @@ class FpEngine:
         fp = accumulate_to
         if fp is None:
             fp = torch.zeros((D, F), dtype=torch.float64, device=self.device)
+
+        stack = getattr(self, "_comp_stack", None)
+        if stack is not None and phiinvs is not None:
+            return self._sweep_stacked(phiinvs, fp, D, F, draw_chunk, batched)
 
         fp_side = None
         main_stream = None
@@ -405,6 +437,40 @@ class FpEngine:
             main_stream.wait_event(ev2)
             fp += fp_side
 
+        return fp[0] if not batched else fp
+
+    def _sweep_stacked(self, phiinvs, fp, D, F, draw_chunk, batched):
+        """Pulsar-batched compressed sweep: one chol + one trsm launch
+        per draw chunk across ALL pulsars."""
+        from fastfp_amd import ops
+
+        st = self._comp_stack
+        P = len(self.blocks)
+        mv = st["mv"]
+        pin = []
+        for i in range(P):
+            p = _t64(phiinvs[i], self.device)
+            p = p[None, :] if p.dim() == 1 else p
+            pin.append(p[:, st["vars"][i]])
+        pinv_var = torch.stack(pin)  # (P, D, mv)
+        fp_pp = torch.empty((P, min(draw_chunk, D), F),
+                            dtype=torch.float64, device=self.device)
+        for lo in range(0, D, draw_chunk):
+            hi = min(lo + draw_chunk, D)
+            phi_var = (
+                1.0 / (pinv_var[:, lo:hi, :] - st["delta0"][:, None, :])
+            ).contiguous()
+            if hi - lo == fp_pp.shape[1]:
+                pp = fp_pp
+            else:  # tail chunk: a dim-1 slice would be non-contiguous
+                pp = torch.empty((P, hi - lo, F), dtype=torch.float64,
+                                 device=self.device)
+            pp.zero_()
+            ops.chol_trsm_fp_accum(
+                st["G"], phi_var, st["K"], st["M0"], st["N0"], pp,
+                gsign=-1.0,
+            )
+            fp[lo:hi] += pp.sum(dim=0)
         return fp[0] if not batched else fp
 
     def _accum_eager(self, blk: PulsarBlock, sigma, fp_out):

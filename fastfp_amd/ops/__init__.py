@@ -171,7 +171,7 @@ def chol_trsm_fp_accum(TNT, phiinv, RHS, sNs, sNr, fp_out, gsign=1.0):
     ``gsign``: +1 direct path (M = sNs - W.W); -1 Schur-compressed draw
     path (M = M0 + W.W) — docs/DESIGN.md §draw compression."""
     ext = _try_load()
-    m = TNT.shape[0]
+    m = TNT.shape[-1]
     mp = check_m(m)
     L, invd = ext.chol_batch(TNT.contiguous(), phiinv.contiguous(), mp)
     ext.trsm_fp_accum(L, invd, RHS, sNs, sNr, fp_out, gsign)

@@ -13,11 +13,11 @@ void launch_sigdots(const double*, const double*, const double*,
 void launch_sbgemm(const double*, const double*, const double*,
                    const double*, int, int, int, int, double*, long, long,
                    int, hipStream_t);
-void launch_chol_batch(const double*, const double*, int, int, int, double*,
-                       double*, hipStream_t);
+void launch_chol_batch(const double*, const double*, int, int, int, int,
+                       double*, double*, hipStream_t);
 void launch_trsm_fp(const double*, const double*, const double*,
-                    const double*, const double*, int, int, int, double,
-                    double*, hipStream_t);
+                    const double*, const double*, int, int, int, int,
+                    double, double*, hipStream_t);
 void launch_blockchol_inv(const double*, const double*, const long*,
                           const long*, const long*, int, double*, double*,
                           hipStream_t);
@@ -78,28 +78,32 @@ void sbgemm(torch::Tensor T, torch::Tensor toas, torch::Tensor ninv,
                 (int)ksplit, stream());
 }
 
-// chol_batch: TNT (m,m), phiinv (D,m) -> L (D,mp,mp), invd (D,mp/16,16,16)
+// chol_batch: TNT (m,m) or (P,m,m), phiinv (D,m) or (P,D,m)
+// -> L (P*D,mp,mp), invd (P*D,mp/16,16,16)
 std::vector<torch::Tensor> chol_batch(torch::Tensor TNT, torch::Tensor phiinv,
                                       int64_t mp) {
   check_f64(TNT, "TNT");
   check_f64(phiinv, "phiinv");
-  const int m = TNT.size(0);
-  const int D = phiinv.size(0);
-  TORCH_CHECK(phiinv.size(1) == m, "phiinv width != m");
+  const bool batched = TNT.dim() == 3;
+  const int P = batched ? TNT.size(0) : 1;
+  const int m = TNT.size(batched ? 1 : 0);
+  const int D = phiinv.size(batched ? 1 : 0);
+  TORCH_CHECK(phiinv.dim() == (batched ? 3 : 2), "phiinv rank");
+  TORCH_CHECK(phiinv.size(batched ? 2 : 1) == m, "phiinv width != m");
   TORCH_CHECK(mp % 16 == 0 && mp <= 128 && mp >= m,
-              "chol_batch requires m <= 128 (basis size per pulsar); "
-              "got m=", m);
+              "chol_batch requires m <= 128 (solve dimension); got m=", m);
   auto opts = TNT.options();
-  auto L = torch::empty({D, mp, mp}, opts);
-  auto invd = torch::empty({D, mp / 16, 16, 16}, opts);
+  auto L = torch::empty({(long)P * D, mp, mp}, opts);
+  auto invd = torch::empty({(long)P * D, mp / 16, 16, 16}, opts);
   launch_chol_batch(TNT.data_ptr<double>(), phiinv.data_ptr<double>(), m,
-                    (int)mp, D, L.data_ptr<double>(),
+                    (int)mp, D, P, L.data_ptr<double>(),
                     invd.data_ptr<double>(), stream());
   return {L, invd};
 }
 
-// trsm_fp_accum: L (D,mp,mp), invd, RHS (mp, 2F+1), sNs (3,F), sNr (2,F),
-// fp (D,F) accumulated in place.
+// trsm_fp_accum: single pulsar (L (D,mp,mp), RHS (mp,2F+1), sNs (3,F),
+// fp (D,F)) or pulsar-batched (L (P*D,mp,mp), RHS (P,mp,2F+1),
+// sNs (P,3,F), fp (P,D,F)); accumulates in place.
 void trsm_fp_accum(torch::Tensor L, torch::Tensor invd, torch::Tensor RHS,
                    torch::Tensor sNs, torch::Tensor sNr, torch::Tensor fp,
                    double gsign) {
@@ -109,15 +113,21 @@ void trsm_fp_accum(torch::Tensor L, torch::Tensor invd, torch::Tensor RHS,
   check_f64(sNs, "sNs");
   check_f64(sNr, "sNr");
   check_f64(fp, "fp");
-  const int D = L.size(0);
+  const bool batched = RHS.dim() == 3;
+  const int P = batched ? RHS.size(0) : 1;
   const int mp = L.size(1);
-  const int F = sNs.size(1);
-  TORCH_CHECK(RHS.size(0) == mp && RHS.size(1) == 2 * F + 1, "RHS shape");
-  TORCH_CHECK(fp.size(0) == D && fp.size(1) == F, "fp shape");
+  const int F = sNs.size(batched ? 2 : 1);
+  const int D = fp.size(batched ? 1 : 0);
+  TORCH_CHECK(L.size(0) == (long)P * D, "L batch != P*D");
+  TORCH_CHECK(RHS.size(batched ? 1 : 0) == mp &&
+                  RHS.size(batched ? 2 : 1) == 2 * F + 1,
+              "RHS shape");
+  TORCH_CHECK(fp.dim() == (batched ? 3 : 2) && fp.size(batched ? 2 : 1) == F,
+              "fp shape");
   TORCH_CHECK(mp % 16 == 0 && mp <= 128, "mp must be <=128, multiple of 16");
   launch_trsm_fp(L.data_ptr<double>(), invd.data_ptr<double>(),
                  RHS.data_ptr<double>(), sNs.data_ptr<double>(),
-                 sNr.data_ptr<double>(), mp, F, D, gsign,
+                 sNr.data_ptr<double>(), mp, F, D, P, gsign,
                  fp.data_ptr<double>(), stream());
 }
 

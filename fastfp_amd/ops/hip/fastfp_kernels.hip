@@ -217,10 +217,16 @@ extern "C" __global__ __launch_bounds__(512) void sbgemm_kernel(
 
 template <int NBT>
 __global__ __launch_bounds__(512, 4) void chol_batch_kernel(
-    const double* __restrict__ TNT /*(m,m)*/,
-    const double* __restrict__ phiinv /*(D,m)*/, int m, int D,
-    double* __restrict__ L /*(D,mp,mp)*/,
-    double* __restrict__ invd /*(D, mp/16, 16, 16)*/) {
+    const double* __restrict__ TNT_all /*(P,m,m)*/,
+    const double* __restrict__ phiinv_all /*(P,D,m)*/, int m, int D,
+    double* __restrict__ L_all /*(P*D,mp,mp)*/,
+    double* __restrict__ invd_all /*(P*D, mp/16, 16, 16)*/) {
+  // grid.y batches PULSARS: one launch factors every (pulsar, draw)
+  const int pp = blockIdx.y;
+  const double* TNT = TNT_all + (long)pp * m * m;
+  const double* phiinv = phiinv_all + (long)pp * D * m;
+  double* L = L_all + (long)pp * D * (NBT * 16) * (NBT * 16);
+  double* invd = invd_all + (long)pp * D * NBT * 256;
   // LDS is sized by the template so small matrices keep multiple
   // workgroups per CU (nb=4: 33 KB -> 4 WG/CU vs one at 134 KB; the
   // serial diagonal phases then overlap ACROSS workgroups -- the PMC
@@ -401,12 +407,15 @@ __global__ __launch_bounds__(512, 4) void chol_batch_kernel(
 
 template <int NBT, int DPG>
 __global__ __launch_bounds__(512, 4) void trsm_fp_kernel(
-    const double* __restrict__ L /*(D,mp,mp)*/,
-    const double* __restrict__ invd /*(D, mp/16, 16, 16)*/,
-    const double* __restrict__ RHS /*(mp, 2F+1)*/,
-    const double* __restrict__ sNs /*(3,F)*/,
-    const double* __restrict__ sNr /*(2,F)*/, int F, int D,
-    double gsign, double* __restrict__ fp /*(D,F)*/) {
+    const double* __restrict__ L_all /*(P*D,mp,mp)*/,
+    const double* __restrict__ invd_all /*(P*D, mp/16, 16, 16)*/,
+    const double* __restrict__ RHS_all /*(P, mp, 2F+1)*/,
+    const double* __restrict__ sNs_all /*(P,3,F)*/,
+    const double* __restrict__ sNr_all /*(P,2,F)*/, int F, int D,
+    double gsign, double* __restrict__ fp_all /*(P,D,F)*/) {
+  // grid.z batches PULSARS (one launch per draw chunk); each pulsar
+  // accumulates into its own fp plane, summed deterministically by the
+  // caller (no cross-pulsar races, no atomics).
   // gsign: +1 for the direct path (M = sNs - W.W), -1 for the
   // Schur-compressed draw path (M = M0 + W.W) -- docs/DESIGN.md.
   // NBT = mp/16 is a template parameter so every W[] index below is
@@ -421,6 +430,13 @@ __global__ __launch_bounds__(512, 4) void trsm_fp_kernel(
   __shared__ double Iv[DPG][NBT][16][17];
   __shared__ double Wu[DPG][NBT * 16];  // the solved u column
 
+  const int pp = blockIdx.z;
+  const double* L = L_all + (long)pp * D * mp * mp;
+  const double* invd = invd_all + (long)pp * D * NBT * 256;
+  const double* RHS = RHS_all + (long)pp * mp * (2L * F + 1);
+  const double* sNs = sNs_all + (long)pp * 3 * F;
+  const double* sNr = sNr_all + (long)pp * 2 * F;
+  double* fp = fp_all + (long)pp * D * F;
   const int d0 = blockIdx.y * DPG;
   const int f0 = blockIdx.x * FPT_FREQS;
   const int tid = threadIdx.x;
@@ -746,10 +762,11 @@ void launch_sbgemm(const double* T, const double* toas, const double* ninv,
 }
 
 void launch_chol_batch(const double* TNT, const double* phiinv, int m, int mp,
-                       int D, double* L, double* invd, hipStream_t stream) {
+                       int D, int P, double* L, double* invd,
+                       hipStream_t stream) {
   // small matrices: 256 threads -> 4 workgroups/CU despite the 115-VGPR
   // diagonal-factor pressure; large: 512 threads for MFMA coverage
-  const dim3 grid(D), blk(mp <= 64 ? 256 : 512);
+  const dim3 grid(D, P), blk(mp <= 64 ? 256 : 512);
   switch (mp >> 4) {
 #define CHOL_CASE(NBT) \
     case NBT: hipLaunchKernelGGL(chol_batch_kernel<NBT>, grid, blk, 0, \
@@ -762,7 +779,8 @@ void launch_chol_batch(const double* TNT, const double* phiinv, int m, int mp,
 
 void launch_trsm_fp(const double* L, const double* invd, const double* RHS,
                     const double* sNs, const double* sNr, int mp, int F,
-                    int D, double gsign, double* fp, hipStream_t stream) {
+                    int D, int P, double gsign, double* fp,
+                    hipStream_t stream) {
   const int ftiles = (F + FPT_FREQS - 1) / FPT_FREQS;
   const dim3 blk(512);
   const int nb = mp >> 4;
@@ -771,7 +789,7 @@ void launch_trsm_fp(const double* L, const double* invd, const double* RHS,
   static const char* dpg_env = getenv("FASTFP_TRSM_DPG");
   static const int dpg_want = dpg_env ? atoi(dpg_env) : 1;  // measured: DPG1 (8 waves/SIMD, 61 VGPR) beats DPG2 (4 waves/SIMD) by ~5%
   if (nb <= 4 && dpg_want >= 2) {
-    const dim3 grid(ftiles, (D + 1) / 2);
+    const dim3 grid(ftiles, (D + 1) / 2, P);
     switch (nb) {
 #define TRSM_CASE2(NBT) \
       case NBT: hipLaunchKernelGGL((trsm_fp_kernel<NBT, 2>), grid, blk, 0, \
@@ -780,7 +798,7 @@ void launch_trsm_fp(const double* L, const double* invd, const double* RHS,
 #undef TRSM_CASE2
     }
   } else {
-    const dim3 grid(ftiles, D);
+    const dim3 grid(ftiles, D, P);
     switch (nb) {
 #define TRSM_CASE(NBT) \
       case NBT: hipLaunchKernelGGL((trsm_fp_kernel<NBT, 1>), grid, blk, 0, \
