@@ -63,3 +63,59 @@ def test_shard_slice_partitions():
                 s = shard_slice(n, r, world)
                 idx.extend(range(n)[s])
             assert idx == list(range(n))
+
+
+def _nmfp_build():
+    from fastfp_amd import get_mats_nmfp, initialize_pta
+    from fastfp_amd.nmfp import NMFp
+
+    psrs = make_synthetic_pta(npsr=2, ntoa=60, ntm=3, seed=0)
+    noise = {"gw_gamma": 13.0 / 3.0, "gw_log10_A": float(np.log10(2e-15))}
+    for p in psrs:
+        noise[f"{p.name}_red_noise_gamma"] = 4.0
+        noise[f"{p.name}_red_noise_log10_A"] = -14.5
+    pta = initialize_pta(psrs, noise, inc_cp=True, rn_comps=3, gwb_comps=3)
+    TNTs, Nvecs, Ts = get_mats_nmfp(pta, noise)
+    D = 5
+    rng = np.random.default_rng(2)
+    samples = {
+        n: (rng.uniform(2, 6, D) if n.endswith("gamma")
+            else rng.uniform(-16, -14, D))
+        for n in pta.params
+    }
+    return psrs, pta, Nvecs, Ts, samples, D
+
+
+def _nmfp_worker(rank, world, port, out_file):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    torch.distributed.init_process_group("gloo", rank=rank, world_size=world)
+    from fastfp_amd.nmfp import NMFp
+
+    psrs, pta, Nvecs, Ts, samples, D = _nmfp_build()
+    my = shard_slice(D, rank, world)
+    local_samples = {k: v[my] for k, v in samples.items()}
+    nm = NMFp(psrs, pta.rn_containers)
+    freqs = np.linspace(4e-9, 5e-8, 6)
+    vals = nm.sweep(freqs, local_samples, Nvecs, Ts, device="cpu")
+    t = torch.as_tensor(vals, dtype=torch.float64)
+    full = all_gather_concat(t, world, dim=0)
+    if rank == 0:
+        np.save(out_file, full.numpy())
+    torch.distributed.destroy_process_group()
+
+
+def test_draw_sharded_nmfp_matches_single(tmp_path):
+    """Draw-sharded + all-gathered NM-Fp == single-process (uneven
+    shards: 5 draws over 2 ranks)."""
+    out_file = str(tmp_path / "nm.npy")
+    mp.spawn(_nmfp_worker, args=(2, 29853, out_file), nprocs=2, join=True)
+    got = np.load(out_file)
+
+    from fastfp_amd.nmfp import NMFp
+
+    psrs, pta, Nvecs, Ts, samples, D = _nmfp_build()
+    nm = NMFp(psrs, pta.rn_containers)
+    freqs = np.linspace(4e-9, 5e-8, 6)
+    want = nm.sweep(freqs, samples, Nvecs, Ts, device="cpu")
+    np.testing.assert_allclose(got, want, rtol=1e-12)

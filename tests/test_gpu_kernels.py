@@ -282,3 +282,25 @@ def test_nmfp_large_m_ecorr_gpu_compressed():
         c.to(DEV)
     gpu = nm.sweep(freqs, samples, Nvecs, Ts, device=DEV)
     np.testing.assert_allclose(gpu, cpu, rtol=1e-7)
+
+
+def test_chol_full_dynamic_range_gpu():
+    """Sigma with the production prior spread: 1e-40 tm bins against
+    ~1e12-1e18 red-noise phi^-1."""
+    ext = _ext()
+    rng = np.random.default_rng(31)
+    ntoa, m, D = 400, 24, 6
+    T = torch.as_tensor(rng.normal(size=(ntoa, m)), dtype=torch.float64, device=DEV)
+    nvec = torch.full((ntoa,), 1e-12, dtype=torch.float64, device=DEV)
+    TNT = (T.T @ (T / nvec[:, None])).contiguous()
+    phiinv = torch.empty((D, m), dtype=torch.float64, device=DEV)
+    phiinv[:, :8] = 1e-40  # improper-flat timing-model prior
+    phiinv[:, 8:] = torch.as_tensor(
+        10.0 ** rng.uniform(12, 18, (D, m - 8)), device=DEV
+    )
+    mp = 32
+    L, invd = ext.chol_batch(TNT, phiinv, mp)
+    want = torch.linalg.cholesky(TNT[None] + torch.diag_embed(phiinv))
+    torch.testing.assert_close(
+        torch.tril(L[:, :m, :m]), want, rtol=1e-9, atol=0.0
+    )
