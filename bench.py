@@ -121,7 +121,7 @@ def main():
     ap.add_argument("--rn-comps", type=int, default=30)
     ap.add_argument("--gwb-comps", type=int, default=30)
     ap.add_argument("--freqs", type=int, default=1000)
-    ap.add_argument("--draws-per-step", type=int, default=500)
+    ap.add_argument("--draws-per-step", type=int, default=1000)
     ap.add_argument("--draw-chunk", type=int, default=512)
     ap.add_argument("--freq-chunk", type=int, default=4096)
     ap.add_argument("--device", type=str, default=None)

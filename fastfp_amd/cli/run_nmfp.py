@@ -145,6 +145,8 @@ def main(
     if world > 1:
         torch.distributed.barrier()
 
+    for cont in pta.rn_containers:
+        cont.to(dev)
     eng = FpEngine(psrs, Nvecs, Ts, device=dev)
     eng.precompute(freqs)
     # Schur draw compression: per-draw solves run at the variable-bin

@@ -76,7 +76,7 @@ class NMFp:
         Ts,
         TNTs=None,
         device: str = None,
-        draw_chunk: int = 32,
+        draw_chunk: int = 512,
         freq_chunk: int = 2048,
         engine: FpEngine = None,
         compress: bool = True,
@@ -101,7 +101,9 @@ class NMFp:
                     [sig.var_slice for sig in self.rn_sigs],
                     [sig.get_phiinv(probe) for sig in self.rn_sigs],
                 )
-        phiinvs = [sig.get_phiinv(samples) for sig in self.rn_sigs]
+        from fastfp_amd.noise import batch_phiinv
+
+        phiinvs = batch_phiinv(self.rn_sigs, samples)
         # scalar-parameter dicts produce (m,) vectors; promote to (1, m)
         phiinvs = [p[None, :] if p.dim() == 1 else p for p in phiinvs]
         if compress and engine.compression_margin(phiinvs) < 1e3:

@@ -519,6 +519,8 @@ __global__ __launch_bounds__(512, 4) void trsm_fp_kernel(
 #pragma unroll
     for (int e = 0; e < DPG; ++e) acc[e] += acc2[e];
     // W[rb] = Iv[rb] * (RHS - sum) = Iv[rb] * (-acc)
+    // (a split even/odd-kk chain costs 7 extra VGPRs -> occupancy
+    // 8 -> 7 waves/SIMD and measures SLOWER; keep the single chain)
     f64x4 sol[DPG];
 #pragma unroll
     for (int e = 0; e < DPG; ++e) sol[e] = f64x4{0, 0, 0, 0};
