@@ -122,7 +122,7 @@ def main():
     ap.add_argument("--gwb-comps", type=int, default=30)
     ap.add_argument("--freqs", type=int, default=1000)
     ap.add_argument("--draws-per-step", type=int, default=1000)
-    ap.add_argument("--draw-chunk", type=int, default=512)
+    ap.add_argument("--draw-chunk", type=int, default=1024)
     ap.add_argument("--freq-chunk", type=int, default=4096)
     ap.add_argument("--device", type=str, default=None)
     ap.add_argument("--no-graph", action="store_true",

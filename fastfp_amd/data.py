@@ -209,13 +209,17 @@ def make_pulsar(
     red_amp: float = 0.0,
     red_gamma: float = 13.0 / 3.0,
     red_ncomps: int = 30,
+    cw_amp: float = 0.0,
+    cw_freq: float = 1e-8,
     rng: np.random.Generator = None,
 ) -> PulsarData:
     """Generate one synthetic pulsar.
 
     Residuals are white noise of std ``toaerr`` per TOA plus (optionally)
     a power-law red-noise realization with amplitude ``red_amp`` drawn on
-    a ``red_ncomps``-component Fourier basis.
+    a ``red_ncomps``-component Fourier basis, plus (optionally) a
+    continuous-wave signal of amplitude ``cw_amp`` (seconds) at
+    ``cw_freq`` (Hz) with a random phase — for Fp detection validation.
     """
     rng = rng or np.random.default_rng(0)
     # epoch-structured TOAs: real PTA observations come in epochs of
@@ -246,6 +250,10 @@ def make_pulsar(
         resid = resid + F @ coeffs
 
     # backend per epoch (one receiver per observation)
+    if cw_amp > 0.0:
+        phase = rng.uniform(0, 2 * np.pi)
+        resid = resid + cw_amp * np.sin(2 * np.pi * cw_freq * toas + phase)
+
     bflags = np.asarray(
         [backends[int(e) % len(backends)] for e in epoch_of], dtype=object
     )
@@ -268,6 +276,8 @@ def make_synthetic_pta(
     ntm: int = 10,
     red_amp: float = 0.0,
     red_gamma: float = 13.0 / 3.0,
+    cw_amp: float = 0.0,
+    cw_freq: float = 1e-8,
     seed: int = 0,
     ragged: bool = True,
 ) -> list:
@@ -292,6 +302,8 @@ def make_synthetic_pta(
                 ntm=ntm,
                 red_amp=red_amp,
                 red_gamma=red_gamma,
+                cw_amp=cw_amp,
+                cw_freq=cw_freq,
                 rng=rng,
             )
         )
