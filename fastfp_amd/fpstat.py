@@ -73,12 +73,31 @@ class FastFp:
         device: str = None,
         freq_chunk: int = 2048,
     ) -> np.ndarray:
-        """Fp over a frequency grid.  Returns (F,) numpy array."""
+        """Fp over a frequency grid.  Returns (F,) numpy array.
+
+        On GPU with a large basis (m > 128, e.g. GP-ECORR models) the
+        direct per-draw solve kernel does not apply; when the PTAModel
+        is available its variable-bin info routes the sweep through the
+        Schur compression automatically (docs/DESIGN.md §7)."""
         if device is None:
             device = "cuda" if torch.cuda.is_available() else "cpu"
         eng = FpEngine(self.psrs, Nvecs, Ts, device=device)
         eng.precompute(freqs, freq_chunk=freq_chunk)
-        fp = eng.sweep(sigmas=sigmas)
+        max_m = max(T.shape[1] for T in Ts)
+        conts = getattr(self.pta, "rn_containers", None)
+        if eng._use_hip and max_m > 128 and conts is not None:
+            phiinvs = [
+                np.diag(np.asarray(sg)) - np.diag(np.asarray(TNT))
+                for sg, TNT in zip(
+                    sigmas, (blk.TNT.cpu().numpy() for blk in eng.blocks)
+                )
+            ]
+            eng.enable_draw_compression(
+                [c.var_slice for c in conts], phiinvs
+            )
+            fp = eng.sweep(phiinvs=phiinvs)
+        else:
+            fp = eng.sweep(sigmas=sigmas)
         return fp.cpu().numpy()
 
 

@@ -304,3 +304,25 @@ def test_chol_full_dynamic_range_gpu():
     torch.testing.assert_close(
         torch.tril(L[:, :m, :m]), want, rtol=1e-9, atol=0.0
     )
+
+
+def test_fastfp_sweep_large_m_gpu():
+    """Plain Fp sweep with m > 128 (GP-ECORR) on GPU auto-routes
+    through the compressed path."""
+    from fastfp_amd import FastFp, get_mats_fp, initialize_pta, make_synthetic_pta
+
+    psrs = make_synthetic_pta(npsr=2, ntoa=700, ntm=5, seed=22)
+    noise = {}
+    for p in psrs:
+        noise[f"{p.name}_red_noise_gamma"] = 4.0
+        noise[f"{p.name}_red_noise_log10_A"] = -14.5
+        for b in np.unique(p.backend_flags):
+            noise[f"{p.name}_basis_ecorr_{b}_log10_ecorr"] = -6.5
+    pta = initialize_pta(psrs, noise, inc_cp=False, rn_comps=6, inc_ecorr=True)
+    Nvecs, Ts, sigmas = get_mats_fp(pta, noise)
+    assert max(T.shape[1] for T in Ts) > 128
+    freqs = np.linspace(4e-9, 5e-8, 11)
+    fp_obj = FastFp(psrs, pta)
+    cpu = fp_obj.sweep(freqs, Nvecs, Ts, sigmas, device="cpu")
+    gpu = fp_obj.sweep(freqs, Nvecs, Ts, sigmas, device=DEV)
+    np.testing.assert_allclose(gpu, cpu, rtol=1e-6)
