@@ -24,7 +24,6 @@ config shape — there is no network for real datasets.
 
 import argparse
 import json
-import os
 import time
 
 import numpy as np

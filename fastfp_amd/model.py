@@ -22,7 +22,6 @@ hard-codes (``/root/reference/fastfp/nmfp.py:247``).  This requires
 from __future__ import annotations
 
 import numpy as np
-import torch
 
 from fastfp_amd.bases import (
     create_freqarray,
