@@ -426,7 +426,7 @@ __global__ __launch_bounds__(512, 4) void trsm_fp_kernel(
   // NBT the solve is short, so two draws share the Lp/Iv staging and
   // barriers, and their MFMA chains interleave on the pipe.
   constexpr int mp = NBT * 16;
-  __shared__ double Lp[DPG][16][NBT * 16 + 1];
+  __shared__ double Lp[2][DPG][16][NBT * 16 + 1];  // double-buffered
   __shared__ double Iv[DPG][NBT][16][17];
   __shared__ double Wu[DPG][NBT * 16];  // the solved u column
 
@@ -479,20 +479,23 @@ __global__ __launch_bounds__(512, 4) void trsm_fp_kernel(
   }
   __syncthreads();  // Iv staged
 
-  // blocked forward substitution, W in registers, DPG draws interleaved
+  // blocked forward substitution, W in registers, DPG draws
+  // interleaved.  The L row-panel is DOUBLE-BUFFERED: each iteration
+  // stages the NEXT row block's panel while computing on the current
+  // one, so there is ONE barrier per row block instead of two and the
+  // staging loads hide under the MFMA phase (the PMC profile charged
+  // ~30% of wave cycles to the staging barriers).
 #pragma unroll
   for (int rb = 0; rb < NBT; ++rb) {
-    if (rb > 0) {
-      __syncthreads();  // prior rb's Lp reads complete
-      const int ncols = rb * 16;
+    if (rb + 1 < NBT) {
+      const int ncols = (rb + 1) * 16;
       for (int idx = tid; idx < ndr * 16 * ncols; idx += 512) {
         const int e = idx / (16 * ncols);
         const int q = idx % (16 * ncols);
         const int r = q / ncols, c = q % ncols;
-        Lp[e][r][c] =
-            L[((long)(d0 + e) * mp + rb * 16 + r) * mp + c];
+        Lp[(rb + 1) & 1][e][r][c] =
+            L[((long)(d0 + e) * mp + (rb + 1) * 16 + r) * mp + c];
       }
-      __syncthreads();
     }
     // two accumulator chains per draw (even/odd cb) deepen the MFMA
     // pipeline; folded together before the diagonal solve
@@ -508,7 +511,7 @@ __global__ __launch_bounds__(512, 4) void trsm_fp_kernel(
       for (int kk = 0; kk < 4; ++kk) {
 #pragma unroll
         for (int e = 0; e < DPG; ++e) {
-          const double a = Lp[e][li][cb * 16 + kk * 4 + lk];
+          const double a = Lp[rb & 1][e][li][cb * 16 + kk * 4 + lk];
           if (cb & 1)
             acc2[e] = MFMA_F64(a, W[e][cb][kk], acc2[e]);
           else
@@ -534,8 +537,8 @@ __global__ __launch_bounds__(512, 4) void trsm_fp_kernel(
     }
 #pragma unroll
     for (int e = 0; e < DPG; ++e) W[e][rb] = sol[e];
+    __syncthreads();  // panel rb+1 staged for all; Lp[rb&1] reads done
   }
-  __syncthreads();
 
   // fused reduction, register-resident: a frequency's sin/cos columns
   // are ADJACENT lanes (li even/odd), so the per-frequency dots come
