@@ -185,6 +185,12 @@ def main(
     )
 
     if rank == 0:
+        if not np.isfinite(full).all():
+            logger.warning(
+                "non-finite Fp values in output (%d of %d) — check the "
+                "noise model / chain for invalid parameters",
+                int(np.sum(~np.isfinite(full))), full.size,
+            )
         with open(os.path.join(outdir, f"{savefile}.npy"), "wb") as f:
             np.save(f, full)
     cleanup()

@@ -153,6 +153,7 @@ class FpEngine:
                     self._precompute_hip(blk, freqs, freq_chunk)
                 else:
                     self._precompute_eager(blk, freqs, freq_chunk)
+        self._stack_direct()
         return self
 
     def _precompute_eager(self, blk: PulsarBlock, freqs, freq_chunk):
@@ -292,6 +293,24 @@ class FpEngine:
         self._stack_compression()
         return self
 
+    def _stack_direct(self):
+        """Stack per-pulsar TNT/RHS/sNs/sNr for single-launch DIRECT
+        sweeps (same dimensions across pulsars; used by the plain-Fp
+        phiinv path and the compression-margin fallback)."""
+        self._direct_stack = None
+        if not self._use_hip:
+            return
+        ms = {blk.m for blk in self.blocks}
+        rs = {tuple(blk.RHS.shape) for blk in self.blocks if blk.RHS is not None}
+        if len(ms) != 1 or len(rs) != 1 or self.blocks[0].m > 128:
+            return
+        self._direct_stack = dict(
+            TNT=torch.stack([b.TNT for b in self.blocks]).contiguous(),
+            RHS=torch.stack([b.RHS for b in self.blocks]).contiguous(),
+            sNs=torch.stack([b.sNs for b in self.blocks]).contiguous(),
+            sNr=torch.stack([b.sNr for b in self.blocks]).contiguous(),
+        )
+
     def _stack_compression(self):
         """When every pulsar compresses to the same variable dimension,
         stack the per-pulsar G/K/M0/N0 so the whole sweep is ONE chol
@@ -373,6 +392,15 @@ class FpEngine:
         stack = getattr(self, "_comp_stack", None)
         if stack is not None and phiinvs is not None:
             return self._sweep_stacked(phiinvs, fp, D, F, draw_chunk, batched)
+        dstack = getattr(self, "_direct_stack", None)
+        if (
+            dstack is not None
+            and phiinvs is not None
+            and all(blk.comp is None for blk in self.blocks)
+        ):
+            return self._sweep_stacked_direct(
+                phiinvs, fp, D, F, draw_chunk, batched
+            )
 
         fp_side = None
         main_stream = None
@@ -469,6 +497,36 @@ class FpEngine:
             ops.chol_trsm_fp_accum(
                 st["G"], phi_var, st["K"], st["M0"], st["N0"], pp,
                 gsign=-1.0,
+            )
+            fp[lo:hi] += pp.sum(dim=0)
+        return fp[0] if not batched else fp
+
+    def _sweep_stacked_direct(self, phiinvs, fp, D, F, draw_chunk, batched):
+        """Pulsar-batched DIRECT sweep (no compression): one chol + one
+        trsm launch per draw chunk on the full-m system."""
+        from fastfp_amd import ops
+
+        st = self._direct_stack
+        P = len(self.blocks)
+        pin = []
+        for i in range(P):
+            p = _t64(phiinvs[i], self.device)
+            p = p[None, :] if p.dim() == 1 else p
+            pin.append(p)
+        pinv_all = torch.stack(pin)  # (P, D, m)
+        fp_pp = torch.empty((P, min(draw_chunk, D), F),
+                            dtype=torch.float64, device=self.device)
+        for lo in range(0, D, draw_chunk):
+            hi = min(lo + draw_chunk, D)
+            if hi - lo == fp_pp.shape[1]:
+                pp = fp_pp
+            else:
+                pp = torch.empty((P, hi - lo, F), dtype=torch.float64,
+                                 device=self.device)
+            pp.zero_()
+            ops.chol_trsm_fp_accum(
+                st["TNT"], pinv_all[:, lo:hi, :].contiguous(), st["RHS"],
+                st["sNs"], st["sNr"], pp, gsign=1.0,
             )
             fp[lo:hi] += pp.sum(dim=0)
         return fp[0] if not batched else fp

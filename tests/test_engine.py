@@ -231,3 +231,37 @@ def test_fp_detects_injected_cw_signal():
     # the chi^2(16) background (mean 8)
     assert abs(peak_f - f_inj) <= (freqs[1] - freqs[0]), (peak_f, f_inj)
     assert fp.max() > 50.0, fp.max()
+
+
+def test_compression_margin_fallback():
+    """Draws with absurdly large phi (phiinv near the jitter floor)
+    must disable compression and still produce the exact direct
+    answer."""
+    psrs = make_synthetic_pta(npsr=2, ntoa=80, ntm=4, seed=15)
+    noise = {"gw_gamma": 13.0 / 3.0, "gw_log10_A": float(np.log10(2e-15))}
+    for p in psrs:
+        noise[f"{p.name}_red_noise_gamma"] = 4.0
+        noise[f"{p.name}_red_noise_log10_A"] = -14.5
+    pta = initialize_pta(psrs, noise, inc_cp=False, rn_comps=4)
+    TNTs, Nvecs, Ts = get_mats_nmfp(pta, noise)
+    D = 3
+    samples = {}
+    for n in pta.params:
+        # gamma ~ 13, log10_A ~ -6: phi astronomically large
+        samples[n] = (np.full(D, 12.9) if n.endswith("gamma")
+                      else np.full(D, -6.0))
+    nm = NMFp(psrs, pta.rn_containers)
+    freqs = np.linspace(4e-9, 5e-8, 4)
+    eng = FpEngine(psrs, Nvecs, Ts, device="cpu")
+    eng.precompute(freqs)
+    eng.enable_draw_compression(
+        [c.var_slice for c in pta.rn_containers],
+        [c.get_phiinv({k: v[0] for k, v in samples.items()})
+         for c in pta.rn_containers],
+    )
+    phiinvs = [c.get_phiinv(samples) for c in pta.rn_containers]
+    assert eng.compression_margin(phiinvs) < 1e3  # the dangerous regime
+    got = nm.sweep(freqs, samples, Nvecs, Ts, device="cpu", engine=None)
+    direct = nm.sweep(freqs, samples, Nvecs, Ts, device="cpu",
+                      compress=False)
+    np.testing.assert_allclose(got, direct, rtol=1e-7)
