@@ -58,6 +58,7 @@ class PTAModel:
         inc_ecorr: bool = False,
         select: str = "backend",
         ecorr_kernel: bool = False,
+        per_psr_tspan: bool = False,
     ):
         """``ecorr_kernel=True`` models ECORR as BLOCK-DIAGONAL white
         noise (enterprise's EcorrKernelNoise) instead of a basis GP:
@@ -67,7 +68,17 @@ class PTAModel:
         (``/root/reference/fastfp/utils.py:30-31``, README.md:22)."""
         if inc_ecorr and ecorr_kernel:
             raise ValueError("choose GP ecorr (inc_ecorr) OR kernel ecorr")
+        if per_psr_tspan and inc_cp:
+            # the reference's Tspan=None + add_curn combination adds the
+            # CURN PSD onto per-pulsar-frequency bins, which mixes
+            # different physical frequencies per pulsar; we reject it
+            # rather than reproduce the quirk (SURVEY.md §2.5 spirit)
+            raise ValueError(
+                "per-pulsar Tspan red-noise bases are incompatible with a "
+                "shared-basis common process; use a PTA-wide Tspan"
+            )
         self.ecorr_kernel = ecorr_kernel
+        self.per_psr_tspan = per_psr_tspan
         if inc_cp:
             assert gwb_comps <= rn_comps, (
                 "shared-basis CURN requires gwb_comps <= rn_comps "
@@ -83,6 +94,8 @@ class PTAModel:
         self.select = select
 
         self.Tspan = get_tspan(psrs)
+        #: PTA-wide red-noise grid (or per-pulsar grids, the reference's
+        #: setup_fp_model(Tspan=None) mode, run_nmfp.py:94-98)
         self.Ffreqs_rn = create_freqarray(self.Tspan, rn_comps)
 
         curn = None
@@ -106,7 +119,12 @@ class PTAModel:
                 ecorr_obj = GPEcorrContainer(psr, weights, fix_wn_vals=self.noise)
                 blocks.append(Uec)
                 n_ec = Uec.shape[1]
-            Frn = fourier_basis(psr.toas, self.Ffreqs_rn)
+            Ffreqs_p = (
+                create_freqarray(psr.Tspan, rn_comps)
+                if per_psr_tspan
+                else self.Ffreqs_rn
+            )
+            Frn = fourier_basis(psr.toas, Ffreqs_p)
             blocks.append(Frn)
             T = np.concatenate(blocks, axis=1)
             self._Ts.append(T)
@@ -130,7 +148,7 @@ class PTAModel:
             self.rn_containers.append(
                 RNContainer(
                     psr,
-                    Ffreqs=self.Ffreqs_rn,
+                    Ffreqs=Ffreqs_p,
                     ncomps=rn_comps,
                     gp_ecorr=inc_ecorr,
                     ecorr_container=ecorr_obj,
@@ -202,6 +220,7 @@ def initialize_pta(
     inc_ecorr=False,
     select="backend",
     ecorr_kernel=False,
+    per_psr_tspan=False,
 ) -> PTAModel:
     """Build the PTA model — signature parity with the reference's
     ``initialize_pta`` (``/root/reference/fastfp/utils.py:104-113``),
@@ -217,6 +236,7 @@ def initialize_pta(
         inc_ecorr=inc_ecorr,
         select=select,
         ecorr_kernel=ecorr_kernel,
+        per_psr_tspan=per_psr_tspan,
     )
 
 

@@ -99,3 +99,33 @@ def test_ecorr_model_builds():
         assert T.shape[1] == p.ntm + n_ec + 6
     for sg in sigmas:
         np.linalg.cholesky(sg)
+
+
+def test_per_psr_tspan_bases():
+    """The reference's setup_fp_model(Tspan=None) mode: red-noise bases
+    on each pulsar's own Tspan (run_nmfp.py:94-98)."""
+    import torch
+
+    from fastfp_amd.bases import create_freqarray
+
+    psrs = make_synthetic_pta(npsr=3, ntoa=70, ntm=3, seed=7)
+    noise = {}
+    for p in psrs:
+        noise[f"{p.name}_red_noise_gamma"] = 4.0
+        noise[f"{p.name}_red_noise_log10_A"] = -14.5
+    pta = initialize_pta(psrs, noise, inc_cp=False, rn_comps=4,
+                         per_psr_tspan=True)
+    for p, cont in zip(psrs, pta.rn_containers):
+        np.testing.assert_allclose(
+            cont.Ffreqs.numpy(), create_freqarray(p.Tspan, 4)
+        )
+    # engine still runs end to end
+    Nvecs, Ts, sigmas = get_mats_fp(pta, noise)
+    from fastfp_amd import FastFp
+
+    fp = FastFp(psrs).sweep(np.linspace(4e-9, 4e-8, 3), Nvecs, Ts,
+                            sigmas, device="cpu")
+    assert np.isfinite(fp).all()
+    # the shared-basis CURN combination is rejected, not silently wrong
+    with pytest.raises(ValueError):
+        initialize_pta(psrs, noise, inc_cp=True, per_psr_tspan=True)
