@@ -50,6 +50,9 @@ def setup_fp_model(psrs, noise, Tspan=None, add_ecorr=False, nrncomps=30,
             rn_comps=nrncomps,
             gwb_comps=ngwbcomps,
             inc_ecorr=add_ecorr,
+            # reference semantics: Tspan=None -> per-pulsar bases
+            # (run_nmfp.py:94-98); a set Tspan -> one shared grid
+            per_psr_tspan=Tspan is None,
         )
     return NMFp(psrs, pta.rn_containers)
 
