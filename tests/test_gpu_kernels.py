@@ -326,3 +326,33 @@ def test_fastfp_sweep_large_m_gpu():
     cpu = fp_obj.sweep(freqs, Nvecs, Ts, sigmas, device="cpu")
     gpu = fp_obj.sweep(freqs, Nvecs, Ts, sigmas, device=DEV)
     np.testing.assert_allclose(gpu, cpu, rtol=1e-6)
+
+
+def test_engine_direct_stacked_gpu_matches_cpu():
+    """Direct (uncompressed) pulsar-stacked sweep on GPU vs CPU eager."""
+    from fastfp_amd import FpEngine, get_mats_nmfp, initialize_pta, make_synthetic_pta
+
+    psrs = make_synthetic_pta(npsr=3, ntoa=400, ntm=4, seed=33, ragged=False)
+    noise = {"gw_gamma": 13.0 / 3.0, "gw_log10_A": float(np.log10(2e-15))}
+    for p in psrs:
+        noise[f"{p.name}_red_noise_gamma"] = 4.0
+        noise[f"{p.name}_red_noise_log10_A"] = -14.5
+    pta = initialize_pta(psrs, noise, inc_cp=True, rn_comps=6, gwb_comps=6)
+    TNTs, Nvecs, Ts = get_mats_nmfp(pta, noise)
+    D = 5
+    rng = np.random.default_rng(3)
+    pars = {
+        n: (rng.uniform(2, 6, D) if n.endswith("gamma")
+            else rng.uniform(-16, -14, D))
+        for n in pta.params
+    }
+    phiinvs = [c.get_phiinv(pars).numpy() for c in pta.rn_containers]
+    freqs = np.linspace(4e-9, 6e-8, 21)
+
+    eng_c = FpEngine(psrs, Nvecs, Ts, device="cpu").precompute(freqs)
+    want = eng_c.sweep(phiinvs=phiinvs).numpy()
+
+    eng_g = FpEngine(psrs, Nvecs, Ts, device=DEV).precompute(freqs)
+    assert eng_g._direct_stack is not None, "stack should engage (same m)"
+    got = eng_g.sweep(phiinvs=phiinvs).cpu().numpy()
+    np.testing.assert_allclose(got, want, rtol=1e-8)
