@@ -65,6 +65,7 @@ def main(
 
     t0 = time.perf_counter()
     Nvecs, Ts, sigmas = get_mats_fp(pta, noise)
+    phiinvs = pta.get_phiinv(noise)
     logger.info(f"Precompute matrix wall time: {time.perf_counter() - t0:.4f} s")
 
     freqs = np.linspace(fmin, fmax, nfreqs)
@@ -73,7 +74,9 @@ def main(
     t0 = time.perf_counter()
     eng = FpEngine(psrs, Nvecs, Ts, device=dev)
     eng.precompute(local)
-    fp_local = eng.sweep(sigmas=sigmas)
+    # the phiinv form lets homogeneous models use the pulsar-stacked
+    # single-launch path; heterogeneous models fall back per pulsar
+    fp_local = eng.sweep(phiinvs=phiinvs)
     fp = all_gather_concat(fp_local.reshape(-1), world).cpu().numpy()
     logger.info(f"Fp-statistic wall time: {time.perf_counter() - t0:.4f} s")
 

@@ -1,0 +1,88 @@
+"""Randomized GPU-vs-CPU sweep validation: random model shapes
+(ragged TOAs, basis sizes across the kernel template range, tiny/odd
+frequency grids, ecorr variants) — run on a GPU box.
+
+    python tools/fuzz_gpu.py [ncases] [seed0]
+"""
+
+import sys
+
+import numpy as np
+
+sys.path.insert(0, ".")
+
+from fastfp_amd import (  # noqa: E402
+    FastFp,
+    get_mats_fp,
+    get_mats_nmfp,
+    initialize_pta,
+    make_synthetic_pta,
+)
+from fastfp_amd.nmfp import NMFp  # noqa: E402
+
+
+def one_case(seed):
+    rng = np.random.default_rng(seed)
+    npsr = int(rng.integers(1, 4))
+    ntoa = int(rng.integers(40, 1200))
+    ntm = int(rng.integers(3, 8))
+    rn = int(rng.integers(1, 9))
+    F = int(rng.integers(1, 150))
+    D = int(rng.integers(1, 40))
+    inc_cp = bool(rng.integers(0, 2))
+    mode = rng.choice(["plain", "gp_ecorr", "kernel_ecorr"])
+    psrs = make_synthetic_pta(npsr=npsr, ntoa=ntoa, ntm=ntm, seed=seed)
+    noise = {"gw_gamma": 13.0 / 3.0, "gw_log10_A": float(np.log10(2e-15))}
+    for p in psrs:
+        noise[f"{p.name}_red_noise_gamma"] = float(rng.uniform(2, 6))
+        noise[f"{p.name}_red_noise_log10_A"] = float(rng.uniform(-16, -14))
+        for b in np.unique(p.backend_flags):
+            noise[f"{p.name}_basis_ecorr_{b}_log10_ecorr"] = float(
+                rng.uniform(-7.5, -6)
+            )
+    pta = initialize_pta(
+        psrs, noise, inc_cp=inc_cp, rn_comps=rn, gwb_comps=max(1, rn - 1),
+        inc_ecorr=(mode == "gp_ecorr"), ecorr_kernel=(mode == "kernel_ecorr"),
+    )
+    desc = (f"seed={seed} npsr={npsr} ntoa={ntoa} ntm={ntm} rn={rn} F={F} "
+            f"D={D} cp={inc_cp} mode={mode}")
+
+    freqs = np.linspace(3e-9, 6e-8, F)
+    TNTs, Nvecs, Ts = get_mats_nmfp(pta, noise)
+    samples = {
+        n: (rng.uniform(2, 6, D) if n.endswith("gamma")
+            else rng.uniform(-16, -14, D))
+        for n in pta.params
+    }
+    nm = NMFp(psrs, pta.rn_containers)
+    cpu = nm.sweep(freqs, samples, Nvecs, Ts, device="cpu")
+    for c in pta.rn_containers:
+        c.to("cuda:0")
+    gpu = nm.sweep(freqs, samples, Nvecs, Ts, device="cuda:0")
+    np.testing.assert_allclose(gpu, cpu, rtol=2e-6, atol=1e-10, err_msg=desc)
+
+    # plain Fp path too
+    Nvecs2, Ts2, sigmas = get_mats_fp(pta, noise)
+    fo = FastFp(psrs, pta)
+    cpu2 = fo.sweep(freqs, Nvecs2, Ts2, sigmas, device="cpu")
+    gpu2 = fo.sweep(freqs, Nvecs2, Ts2, sigmas, device="cuda:0")
+    np.testing.assert_allclose(gpu2, cpu2, rtol=2e-6, atol=1e-10, err_msg=desc)
+    return desc
+
+
+def main():
+    n = int(sys.argv[1]) if len(sys.argv) > 1 else 20
+    s0 = int(sys.argv[2]) if len(sys.argv) > 2 else 1000
+    bad = 0
+    for k in range(n):
+        try:
+            print("OK ", one_case(s0 + k), flush=True)
+        except AssertionError as e:
+            bad += 1
+            print("FAIL", str(e)[:400], flush=True)
+    print(f"{n - bad}/{n} cases passed")
+    sys.exit(1 if bad else 0)
+
+
+if __name__ == "__main__":
+    main()
