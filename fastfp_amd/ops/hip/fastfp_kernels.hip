@@ -789,8 +789,9 @@ void launch_trsm_fp(const double* L, const double* invd, const double* RHS,
   const int ftiles = (F + FPT_FREQS - 1) / FPT_FREQS;
   const dim3 blk(512);
   const int nb = mp >> 4;
-  // two draws per workgroup when the solve is small (compressed path);
-  // FASTFP_TRSM_DPG=1 selects the single-draw variant (A/B tuning)
+  // DPG selects draws per workgroup for small solves; the measured
+  // default is 1 (8 waves/SIMD at 61 VGPR beats DPG=2's 2 chains at
+  // 4 waves/SIMD by ~5%); FASTFP_TRSM_DPG=2 re-enables the A/B arm
   static const char* dpg_env = getenv("FASTFP_TRSM_DPG");
   static const int dpg_want = dpg_env ? atoi(dpg_env) : 1;  // measured: DPG1 (8 waves/SIMD, 61 VGPR) beats DPG2 (4 waves/SIMD) by ~5%
   if (nb <= 4 && dpg_want >= 2) {
