@@ -394,11 +394,12 @@ __global__ __launch_bounds__(512, 4) void chol_batch_kernel(
 // runs with W held in registers (acc layout), zero cross-lane moves:
 //   acc  = -RHS_rb;  acc += sum_cb L[rb,cb] (x) W[cb]   (a from LDS Lp)
 //   W[rb] = Iv[rb] (x) (-acc)                           (a from LDS Iv)
-// LDS holds only the L row-panel, the inverted diagonal blocks, and a
-// 16-row staging buffer for the fused reduction -> ~50 KB, 2 WG/CU,
-// vs 151 KB / 1 WG/CU for the LDS-resident W variant
-// (profiles/r01_initial_stats.md).
-// grid.x = ceil(F / 63), grid.y = D;  block = 512 (8 waves)
+// The reduction is also register/shfl-resident (a frequency's sin/cos
+// columns are ADJACENT lanes); LDS holds only the double-buffered L
+// row-panel, the inverted diagonal blocks and the tiny solved-u column
+// (~18 KB at the compressed shape -> 4 WG/CU, occupancy bounded by the
+// 61-VGPR budget at 8 waves/SIMD).
+// grid = (ceil(F/63), ceil(D/DPG), P);  block = 512 (8 waves)
 // cols layout: [s0 c0 s1 c1 ... s62 c62 | u | pad]
 // ---------------------------------------------------------------------
 #define FPT_COLS 128
