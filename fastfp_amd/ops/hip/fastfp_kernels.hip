@@ -280,8 +280,14 @@ __global__ __launch_bounds__(512, 4) void chol_batch_kernel(
         A_(dk0 + i, dk0 + c) = row[c];                                       \
     const int c = i;                                                         \
     double diag[16], x[16];                                                  \
+    /* lane i computes only 1/L_ii; the rest arrive by shfl (16x less   */   \
+    /* serial f64 division work than every lane inverting all 16).      */   \
+    /* predicated select avoids a dynamic register index (scratch!)     */   \
+    double dii = 0.0;                                                        \
+    _Pragma("unroll") for (int r = 0; r < 16; ++r) if (r == i) dii = row[r]; \
+    const double myrcp = 1.0 / dii;                                          \
     _Pragma("unroll") for (int r = 0; r < 16; ++r) diag[r] =                 \
-        1.0 / __shfl(row[r], r, 16);                                         \
+        __shfl(myrcp, r, 16);                                                \
     _Pragma("unroll") for (int r = 0; r < 16; ++r) {                         \
       double acc2 = 0.0;                                                     \
       _Pragma("unroll") for (int t = 0; t < 16; ++t) {                       \
@@ -655,9 +661,16 @@ extern "C" __global__ __launch_bounds__(512) void blockchol_inv_kernel(
   // X = L^{-1}: lane holds column c = i
   const int c = i;
   double diag[BLK_MAX], x[BLK_MAX];
+  // one reciprocal per lane, broadcast by shfl (predicated select
+  // avoids a dynamic register index)
+  double dii = 1.0;
 #pragma unroll
   for (int r = 0; r < BLK_MAX; ++r)
-    diag[r] = (r < sz) ? 1.0 / __shfl(row[r], r, 32) : 0.0;
+    if (r == i && r < sz) dii = row[r];
+  const double myrcp = (i < sz) ? 1.0 / dii : 0.0;
+#pragma unroll
+  for (int r = 0; r < BLK_MAX; ++r)
+    diag[r] = (r < sz) ? __shfl(myrcp, r, 32) : 0.0;
 #pragma unroll
   for (int r = 0; r < BLK_MAX; ++r) {
     double acc = 0.0;
