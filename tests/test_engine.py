@@ -265,3 +265,27 @@ def test_compression_margin_fallback():
     direct = nm.sweep(freqs, samples, Nvecs, Ts, device="cpu",
                       compress=False)
     np.testing.assert_allclose(got, direct, rtol=1e-7)
+
+
+def test_sweep_bitwise_invariant_to_draw_chunk():
+    """Chunking the draw axis must not change results AT ALL (each
+    chunk's accumulation order is per-element identical)."""
+    psrs = make_synthetic_pta(npsr=2, ntoa=70, ntm=3, seed=16)
+    noise = {"gw_gamma": 13.0 / 3.0, "gw_log10_A": float(np.log10(2e-15))}
+    for p in psrs:
+        noise[f"{p.name}_red_noise_gamma"] = 4.0
+        noise[f"{p.name}_red_noise_log10_A"] = -14.5
+    pta = initialize_pta(psrs, noise, inc_cp=True, rn_comps=4, gwb_comps=3)
+    TNTs, Nvecs, Ts = get_mats_nmfp(pta, noise)
+    D = 7
+    rng = np.random.default_rng(4)
+    samples = {
+        n: (rng.uniform(2, 6, D) if n.endswith("gamma")
+            else rng.uniform(-16, -14, D))
+        for n in pta.params
+    }
+    nm = NMFp(psrs, pta.rn_containers)
+    freqs = np.linspace(4e-9, 5e-8, 5)
+    a = nm.sweep(freqs, samples, Nvecs, Ts, device="cpu", draw_chunk=2)
+    b = nm.sweep(freqs, samples, Nvecs, Ts, device="cpu", draw_chunk=7)
+    np.testing.assert_array_equal(a, b)

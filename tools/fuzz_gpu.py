@@ -30,7 +30,7 @@ def one_case(seed):
     F = int(rng.integers(1, 150))
     D = int(rng.integers(1, 40))
     inc_cp = bool(rng.integers(0, 2))
-    mode = rng.choice(["plain", "gp_ecorr", "kernel_ecorr"])
+    mode = rng.choice(["plain", "gp_ecorr", "kernel_ecorr", "per_psr"])
     psrs = make_synthetic_pta(npsr=npsr, ntoa=ntoa, ntm=ntm, seed=seed)
     noise = {"gw_gamma": 13.0 / 3.0, "gw_log10_A": float(np.log10(2e-15))}
     for p in psrs:
@@ -41,8 +41,10 @@ def one_case(seed):
                 rng.uniform(-7.5, -6)
             )
     pta = initialize_pta(
-        psrs, noise, inc_cp=inc_cp, rn_comps=rn, gwb_comps=max(1, rn - 1),
+        psrs, noise, inc_cp=inc_cp and mode != "per_psr",
+        rn_comps=rn, gwb_comps=max(1, rn - 1),
         inc_ecorr=(mode == "gp_ecorr"), ecorr_kernel=(mode == "kernel_ecorr"),
+        per_psr_tspan=(mode == "per_psr"),
     )
     desc = (f"seed={seed} npsr={npsr} ntoa={ntoa} ntm={ntm} rn={rn} F={F} "
             f"D={D} cp={inc_cp} mode={mode}")
