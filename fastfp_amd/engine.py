@@ -355,6 +355,7 @@ class FpEngine:
     def disable_draw_compression(self):
         for blk in self.blocks:
             blk.comp = None
+        self._comp_stack = None  # the stacked path must also fall back
         return self
 
     # ------------------------------------------------------------------

@@ -289,3 +289,25 @@ def test_sweep_bitwise_invariant_to_draw_chunk():
     a = nm.sweep(freqs, samples, Nvecs, Ts, device="cpu", draw_chunk=2)
     b = nm.sweep(freqs, samples, Nvecs, Ts, device="cpu", draw_chunk=7)
     np.testing.assert_array_equal(a, b)
+
+
+def test_disable_compression_clears_stack():
+    """disable_draw_compression must clear BOTH the per-pulsar state and
+    the stacked-launch cache (the GPU sweep consults the stack first)."""
+    psrs = make_synthetic_pta(npsr=2, ntoa=60, ntm=3, seed=18)
+    noise = {"gw_gamma": 13.0 / 3.0, "gw_log10_A": float(np.log10(2e-15))}
+    for p in psrs:
+        noise[f"{p.name}_red_noise_gamma"] = 4.0
+        noise[f"{p.name}_red_noise_log10_A"] = -14.5
+    pta = initialize_pta(psrs, noise, inc_cp=False, rn_comps=3)
+    TNTs, Nvecs, Ts = get_mats_nmfp(pta, noise)
+    eng = FpEngine(psrs, Nvecs, Ts, device="cpu")
+    eng.precompute(np.linspace(4e-9, 5e-8, 4))
+    eng.enable_draw_compression(
+        [c.var_slice for c in pta.rn_containers],
+        [c.get_phiinv(noise) for c in pta.rn_containers],
+    )
+    eng._comp_stack = object()  # simulate the GPU stacked cache
+    eng.disable_draw_compression()
+    assert eng._comp_stack is None
+    assert all(blk.comp is None for blk in eng.blocks)
