@@ -147,6 +147,9 @@ class FpEngine:
         freqs = _t64(freqs, self.device).reshape(-1)
         self.freqs = freqs
         F = freqs.shape[0]
+        # any existing compression state was built against the OLD
+        # frequency grid; callers re-enable after re-precomputing
+        self.disable_draw_compression()
         with _roctx("fastfp:freq_precompute"):
             for blk in self.blocks:
                 if self._use_hip:
