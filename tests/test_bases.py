@@ -1,7 +1,6 @@
 """Basis construction tests."""
 
 import numpy as np
-import pytest
 
 from fastfp_amd.bases import (
     create_freqarray,

@@ -7,7 +7,6 @@ import pytest
 
 from fastfp_amd import (
     FastFp,
-    FpEngine,
     get_mats_fp,
     initialize_pta,
     make_synthetic_pta,

@@ -5,7 +5,6 @@ sharded + all-gathered spectrum must equal the single-process result
 import os
 
 import numpy as np
-import pytest
 import torch
 import torch.multiprocessing as mp
 

@@ -3,7 +3,6 @@ the reference-shaped parity path."""
 
 import numpy as np
 import pytest
-import torch
 
 from fastfp_amd import (
     FastFp,

@@ -104,8 +104,6 @@ def test_ecorr_model_builds():
 def test_per_psr_tspan_bases():
     """The reference's setup_fp_model(Tspan=None) mode: red-noise bases
     on each pulsar's own Tspan (run_nmfp.py:94-98)."""
-    import torch
-
     from fastfp_amd.bases import create_freqarray
 
     psrs = make_synthetic_pta(npsr=3, ntoa=70, ntm=3, seed=7)

@@ -3,7 +3,6 @@
 
 import numpy as np
 import pytest
-import torch
 
 from fastfp_amd.bases import create_freqarray
 from fastfp_amd.constants import fyr

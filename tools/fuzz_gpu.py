@@ -9,7 +9,8 @@ import sys
 
 import numpy as np
 
-sys.path.insert(0, ".")
+sys.path.insert(0, __import__("os").path.abspath(
+    __import__("os").path.join(__import__("os").path.dirname(__file__), "..")))
 
 from fastfp_amd import (  # noqa: E402
     FastFp,

@@ -9,7 +9,8 @@ import sys
 import numpy as np
 import torch
 
-sys.path.insert(0, ".")
+sys.path.insert(0, __import__("os").path.abspath(
+    __import__("os").path.join(__import__("os").path.dirname(__file__), "..")))
 from fastfp_amd.ops import _fastfp_hip as ext  # noqa: E402
 
 DEV = "cuda:0"

@@ -9,7 +9,8 @@ import time
 import numpy as np
 import torch
 
-sys.path.insert(0, ".")
+sys.path.insert(0, __import__("os").path.abspath(
+    __import__("os").path.join(__import__("os").path.dirname(__file__), "..")))
 from fastfp_amd import FpEngine, get_mats_nmfp, initialize_pta, make_synthetic_pta  # noqa: E402
 from fastfp_amd.noise import batch_phiinv, check_batch_homogeneous  # noqa: E402
 
