@@ -310,3 +310,24 @@ def test_disable_compression_clears_stack():
     eng.disable_draw_compression()
     assert eng._comp_stack is None
     assert all(blk.comp is None for blk in eng.blocks)
+
+
+def test_nmfp_at_fixed_params_equals_plain_fp():
+    """NM-Fp rows evaluated AT the noise-dict parameters must equal the
+    plain Fp sweep (cross-validates the two API paths end to end)."""
+    psrs = make_synthetic_pta(npsr=3, ntoa=80, ntm=3, seed=19)
+    noise = {"gw_gamma": 13.0 / 3.0, "gw_log10_A": float(np.log10(2e-15))}
+    for p in psrs:
+        noise[f"{p.name}_red_noise_gamma"] = 4.2
+        noise[f"{p.name}_red_noise_log10_A"] = -14.3
+    pta = initialize_pta(psrs, noise, inc_cp=True, rn_comps=4, gwb_comps=3)
+    Nvecs, Ts, sigmas = get_mats_fp(pta, noise)
+    freqs = np.linspace(4e-9, 5e-8, 6)
+    plain = FastFp(psrs, pta).sweep(freqs, Nvecs, Ts, sigmas, device="cpu")
+
+    D = 3
+    samples = {k: np.full(D, float(noise[k])) for k in pta.params}
+    nm = NMFp(psrs, pta.rn_containers)
+    vals = nm.sweep(freqs, samples, Nvecs, Ts, device="cpu")
+    for d in range(D):
+        np.testing.assert_allclose(vals[d], plain, rtol=1e-7)
