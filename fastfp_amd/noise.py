@@ -60,20 +60,20 @@ def white_noise_nvec(psr, noise: dict = None, simple_wn: bool = True,
     """
     if simple_wn:
         return psr.toaerrs**2
+    noise = noise or {}
     nvec = np.zeros(psr.ntoa, dtype=np.float64)
-    keysel = (
-        np.unique(psr.backend_flags)
-        if select == "backend"
-        else np.array(["all"], dtype=object)
-    )
-    for b in keysel:
-        mask = (
-            np.asarray(psr.backend_flags == b)
-            if select == "backend"
-            else np.ones(psr.ntoa, dtype=bool)
-        )
-        efac = float(noise.get(f"{psr.name}_{b}_efac", 1.0)) if noise else 1.0
-        l10eq = noise.get(f"{psr.name}_{b}_log10_t2equad", None) if noise else None
+    if select == "backend":
+        groups = [
+            (np.asarray(psr.backend_flags == b), f"{psr.name}_{b}")
+            for b in np.unique(psr.backend_flags)
+        ]
+    else:
+        # no selection: one parameter set per pulsar, enterprise's
+        # un-selected naming ("{psr}_efac")
+        groups = [(np.ones(psr.ntoa, dtype=bool), psr.name)]
+    for mask, prefix in groups:
+        efac = float(noise.get(f"{prefix}_efac", 1.0))
+        l10eq = noise.get(f"{prefix}_log10_t2equad", None)
         equad2 = 10.0 ** (2.0 * float(l10eq)) if l10eq is not None else 0.0
         nvec[mask] = efac**2 * (psr.toaerrs[mask] ** 2 + equad2)
     return nvec

@@ -139,3 +139,12 @@ def test_white_noise_nvec_backends(psr):
     full = white_noise_nvec(psr, noise, simple_wn=False)
     want = 1.5**2 * (psr.toaerrs**2 + 1e-12)
     np.testing.assert_allclose(full, want, rtol=1e-12)
+
+
+def test_white_noise_no_selection(psr):
+    """select != backend: one parameter set per pulsar with the
+    un-selected enterprise naming ({psr}_efac)."""
+    noise = {f"{psr.name}_efac": 1.2, f"{psr.name}_log10_t2equad": -6.5}
+    nv = white_noise_nvec(psr, noise, simple_wn=False, select="none")
+    want = 1.2**2 * (psr.toaerrs**2 + 10.0 ** (2 * -6.5))
+    np.testing.assert_allclose(nv, want, rtol=1e-12)
