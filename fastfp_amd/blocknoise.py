@@ -121,6 +121,37 @@ class BlockNoise:
         return self._t[key]
 
     # ------------------------------------------------------------------
+    def _size_groups(self, device, dtype):
+        """Cached per-(device) grouped gather indices and stacked dense
+        inverse blocks for the batched solve."""
+        key = (str(device), str(dtype))
+        cache = getattr(self, "_groups", None)
+        if cache is None:
+            cache = self._groups = {}
+        if key not in cache:
+            groups = []
+            for s in np.unique(self.sizes):
+                sel = np.nonzero(self.sizes == s)[0]
+                offs = self.offsets[sel]
+                idx = torch.as_tensor(
+                    (offs[:, None] + np.arange(s)[None, :]).ravel(),
+                    device=device,
+                )
+                invs = torch.as_tensor(
+                    np.stack(
+                        [
+                            self.inv_packed[self.poff[b] : self.poff[b] + s * s]
+                            .reshape(s, s)
+                            for b in sel
+                        ]
+                    ),
+                    device=device,
+                    dtype=dtype,
+                )
+                groups.append((int(s), len(sel), idx, invs))
+            cache[key] = groups
+        return cache[key]
+
     def solve(self, X):
         """``N^{-1} X`` for X of shape (ntoa,) or (ntoa, k), in the
         PERMUTED TOA order.  Torch or numpy in, same type out."""
@@ -130,27 +161,9 @@ class BlockNoise:
         if vec:
             Xt = Xt[:, None]
         out = torch.empty_like(Xt)
-        # group blocks by size for batched application
-        for s in np.unique(self.sizes):
-            sel = np.nonzero(self.sizes == s)[0]
-            offs = self.offsets[sel]
-            idx = torch.as_tensor(
-                (offs[:, None] + np.arange(s)[None, :]).ravel(),
-                device=Xt.device,
-            )
-            invs = torch.as_tensor(
-                np.stack(
-                    [
-                        self.inv_packed[self.poff[b] : self.poff[b] + s * s]
-                        .reshape(s, s)
-                        for b in sel
-                    ]
-                ),
-                device=Xt.device,
-                dtype=Xt.dtype,
-            )
-            xb = Xt[idx].reshape(len(sel), s, -1)
-            out[idx] = torch.bmm(invs, xb).reshape(len(sel) * s, -1)
+        for s, nblk, idx, invs in self._size_groups(Xt.device, Xt.dtype):
+            xb = Xt[idx].reshape(nblk, s, -1)
+            out[idx] = torch.bmm(invs, xb).reshape(nblk * s, -1)
         out = out[:, 0] if vec else out
         return out.numpy() if is_np else out
 
