@@ -356,3 +356,60 @@ def test_engine_direct_stacked_gpu_matches_cpu():
     assert eng_g._direct_stack is not None, "stack should engage (same m)"
     got = eng_g.sweep(phiinvs=phiinvs).cpu().numpy()
     np.testing.assert_allclose(got, want, rtol=1e-8)
+
+
+def test_graph_captured_sweep_bitwise_equals_eager():
+    """A hipGraph-captured sweep (the bench's step structure) must
+    reproduce the eager sweep BITWISE on replay."""
+    from fastfp_amd import FpEngine, get_mats_nmfp, initialize_pta, make_synthetic_pta
+    from fastfp_amd.noise import batch_phiinv, check_batch_homogeneous
+
+    psrs = make_synthetic_pta(npsr=3, ntoa=400, ntm=5, seed=41)
+    noise = {"gw_gamma": 13.0 / 3.0, "gw_log10_A": float(np.log10(2e-15))}
+    for p in psrs:
+        noise[f"{p.name}_red_noise_gamma"] = 4.0
+        noise[f"{p.name}_red_noise_log10_A"] = -14.5
+    pta = initialize_pta(psrs, noise, inc_cp=True, rn_comps=5, gwb_comps=5)
+    TNTs, Nvecs, Ts = get_mats_nmfp(pta, noise)
+    for c in pta.rn_containers:
+        c.to(DEV)
+    eng = FpEngine(psrs, Nvecs, Ts, device=DEV)
+    freqs = np.linspace(4e-9, 6e-8, 40)
+    eng.precompute(freqs)
+    eng.enable_draw_compression(
+        [c.var_slice for c in pta.rn_containers],
+        [c.get_phiinv(noise) for c in pta.rn_containers],
+    )
+    D = 12
+    rng = np.random.default_rng(1)
+    pool = {
+        n: torch.as_tensor(
+            rng.uniform(2, 6, D) if n.endswith("gamma")
+            else rng.uniform(-16, -14, D),
+            dtype=torch.float64, device=DEV,
+        )
+        for n in pta.params
+    }
+    homog = check_batch_homogeneous(pta.rn_containers)
+    fp_accum = torch.zeros((D, 40), dtype=torch.float64, device=DEV)
+
+    def step():
+        phiinvs = batch_phiinv(pta.rn_containers, pool, homogeneous=homog)
+        fp_accum.zero_()
+        eng.sweep(phiinvs=phiinvs, draw_chunk=8, accumulate_to=fp_accum)
+
+    step()
+    torch.cuda.synchronize()
+    eager = fp_accum.cpu().numpy().copy()
+
+    side = torch.cuda.Stream()
+    side.wait_stream(torch.cuda.current_stream())
+    with torch.cuda.stream(side):
+        step()
+    torch.cuda.current_stream().wait_stream(side)
+    g = torch.cuda.CUDAGraph()
+    with torch.cuda.graph(g):
+        step()
+    g.replay()
+    torch.cuda.synchronize()
+    np.testing.assert_array_equal(fp_accum.cpu().numpy(), eager)
