@@ -24,11 +24,13 @@ from fastfp_amd.nmfp import NMFp  # noqa: E402
 
 def one_case(seed):
     rng = np.random.default_rng(seed)
-    npsr = int(rng.integers(1, 4))
-    ntoa = int(rng.integers(40, 1200))
-    ntm = int(rng.integers(3, 8))
-    rn = int(rng.integers(1, 9))
-    F = int(rng.integers(1, 150))
+    npsr = int(rng.integers(1, 6))
+    ntoa = int(rng.integers(40, 4000))
+    ntm = int(rng.integers(3, 10))
+    # up to 40 components: covers the compressed solve at NBT=5 (the
+    # substitution-kernel dispatch) as well as the small-NBT DPG path
+    rn = int(rng.integers(1, 41))
+    F = int(rng.integers(1, 300))
     D = int(rng.integers(1, 40))
     inc_cp = bool(rng.integers(0, 2))
     mode = rng.choice(["plain", "gp_ecorr", "kernel_ecorr", "per_psr"])
