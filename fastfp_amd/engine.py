@@ -241,6 +241,19 @@ class FpEngine:
             pf[sl] = delta0
             sigma0 = blk.TNT + torch.diag(pf)
             L0 = torch.linalg.cholesky(sigma0)
+            # rank-deficient TNT (e.g. basis larger than the TOA count)
+            # leaves Sigma_0 supported only by the jitter along its null
+            # space — the compressed correction then amplifies roundoff.
+            # Null-space Cholesky pivots land AT the jitter scale
+            # (ld^2 ~ delta), healthy pivots far above it; such pulsars
+            # stay on the exact direct path (the sweep supports mixed
+            # per-pulsar compression).
+            ld = torch.diagonal(L0)
+            if blk.ntoa < m or float((ld * ld).min()) < 1e3 * float(
+                delta0.max()
+            ):
+                blk.comp = None
+                continue
             RHSe = blk.RHS[:m, :]  # (m, 2F+1)
             ncols = RHSe.shape[1]
             mv = len(range(*sl.indices(m)))
