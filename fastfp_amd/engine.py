@@ -241,17 +241,13 @@ class FpEngine:
             pf[sl] = delta0
             sigma0 = blk.TNT + torch.diag(pf)
             L0 = torch.linalg.cholesky(sigma0)
-            # rank-deficient TNT (e.g. basis larger than the TOA count)
-            # leaves Sigma_0 supported only by the jitter along its null
-            # space — the compressed correction then amplifies roundoff.
-            # Null-space Cholesky pivots land AT the jitter scale
-            # (ld^2 ~ delta), healthy pivots far above it; such pulsars
-            # stay on the exact direct path (the sweep supports mixed
-            # per-pulsar compression).
-            ld = torch.diagonal(L0)
-            if blk.ntoa < m or float((ld * ld).min()) < 1e3 * float(
-                delta0.max()
-            ):
+            # rank-deficient TNT (basis larger than the TOA count)
+            # leaves Sigma_0 supported only by the jitter along its
+            # null space — keep such pulsars on the exact direct path.
+            # (Near-degeneracy SHORT of deficiency is handled by the
+            # empirical probe below: Cholesky-pivot heuristics measured
+            # uncorrelated with the actual compression error.)
+            if blk.ntoa < m:
                 blk.comp = None
                 continue
             RHSe = blk.RHS[:m, :]  # (m, 2F+1)
@@ -306,8 +302,55 @@ class FpEngine:
                 G=G, K=K.contiguous(), M0=M0, N0=N0, var=sl, mv=mv,
                 delta0=delta0,
             )
+        self._probe_compression(phiinv_fixed)
         self._stack_compression()
         return self
+
+    def _probe_compression(self, phiinv_fixed, tol: float = 1e-5):
+        """Empirical accuracy guard: evaluate one probe draw per pulsar
+        through BOTH the compressed and the direct path and drop
+        compression where they disagree beyond ``tol`` — conditioning
+        heuristics (Cholesky pivot ratios vs the jitter) measured
+        uncorrelated with the real error, so measure instead."""
+        F = self.freqs.shape[0]
+        for blk, pf in zip(self.blocks, phiinv_fixed):
+            if blk.comp is None:
+                continue
+            if blk.m > 128 and self._use_hip:
+                continue  # no direct GPU reference; compression required
+            pinv = _t64(pf, self.device).reshape(1, -1)
+            c = blk.comp
+            fpA = torch.zeros((1, F), dtype=torch.float64, device=self.device)
+            fpB = torch.zeros_like(fpA)
+            phi_var = (1.0 / (pinv[:, c["var"]] - c["delta0"][None, :])).contiguous()
+            try:
+                if self._use_hip:
+                    from fastfp_amd import ops
+
+                    ops.chol_trsm_fp_accum(
+                        c["G"], phi_var, c["K"], c["M0"], c["N0"], fpA,
+                        gsign=-1.0,
+                    )
+                    ops.chol_trsm_fp_accum(
+                        blk.TNT, pinv.contiguous(), blk.RHS, blk.sNs,
+                        blk.sNr, fpB, gsign=1.0,
+                    )
+                else:
+                    sigc = c["G"][None, :, :] + torch.diag_embed(phi_var)
+                    self._accum_eager_mats(
+                        sigc, c["K"], c["M0"], c["N0"], fpA, -1.0
+                    )
+                    sigma = blk.TNT[None, :, :] + torch.diag_embed(pinv)
+                    self._accum_eager(blk, sigma, fpB)
+            except torch.linalg.LinAlgError:
+                blk.comp = None  # pathological probe (e.g. phi < jitter)
+                continue
+            scale = fpB.abs().max().clamp_min(1e-30)
+            err = ((fpA - fpB).abs() / (fpB.abs() + 1e-3 * scale)).max()
+            # NaN-safe: a non-finite probe (GPU kernels don't raise) or
+            # a too-large error both disqualify the compressed path
+            if not bool(torch.isfinite(err)) or float(err) > tol:
+                blk.comp = None
 
     def _stack_direct(self):
         """Stack per-pulsar TNT/RHS/sNs/sNr for single-launch DIRECT

@@ -258,8 +258,11 @@ def test_compression_margin_fallback():
         [c.get_phiinv({k: v[0] for k, v in samples.items()})
          for c in pta.rn_containers],
     )
-    phiinvs = [c.get_phiinv(samples) for c in pta.rn_containers]
-    assert eng.compression_margin(phiinvs) < 1e3  # the dangerous regime
+    # first guard layer: the enable-time accuracy probe already rejects
+    # compression at these pathological parameters
+    assert all(blk.comp is None for blk in eng.blocks)
+    # and the end-to-end sweep (second layer: the per-call margin check)
+    # still produces the exact direct answer
     got = nm.sweep(freqs, samples, Nvecs, Ts, device="cpu", engine=None)
     direct = nm.sweep(freqs, samples, Nvecs, Ts, device="cpu",
                       compress=False)
