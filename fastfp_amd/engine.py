@@ -345,8 +345,11 @@ class FpEngine:
             except torch.linalg.LinAlgError:
                 blk.comp = None  # pathological probe (e.g. phi < jitter)
                 continue
+            # error against the SPECTRUM scale: per-frequency relative
+            # error is meaningless at cancellation-dominated frequencies
+            # (docs/DESIGN.md §8) and mis-flags healthy models
             scale = fpB.abs().max().clamp_min(1e-30)
-            err = ((fpA - fpB).abs() / (fpB.abs() + 1e-3 * scale)).max()
+            err = (fpA - fpB).abs().max() / scale
             # NaN-safe: a non-finite probe (GPU kernels don't raise) or
             # a too-large error both disqualify the compressed path
             if not bool(torch.isfinite(err)) or float(err) > tol:

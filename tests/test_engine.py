@@ -334,3 +334,24 @@ def test_nmfp_at_fixed_params_equals_plain_fp():
     vals = nm.sweep(freqs, samples, Nvecs, Ts, device="cpu")
     for d in range(D):
         np.testing.assert_allclose(vals[d], plain, rtol=1e-7)
+
+
+def test_probe_keeps_compression_on_healthy_model():
+    """The enable-time accuracy probe must NOT disable compression for
+    a healthy (bench-like) model — regression guard for the probe's
+    error metric."""
+    psrs = make_synthetic_pta(npsr=3, ntoa=400, ntm=8, seed=77)
+    noise = {"gw_gamma": 13.0 / 3.0, "gw_log10_A": float(np.log10(2e-15))}
+    for p in psrs:
+        noise[f"{p.name}_red_noise_gamma"] = 13.0 / 3.0
+        noise[f"{p.name}_red_noise_log10_A"] = -14.5
+    pta = initialize_pta(psrs, noise, inc_cp=True, rn_comps=8, gwb_comps=8)
+    TNTs, Nvecs, Ts = get_mats_nmfp(pta, noise)
+    eng = FpEngine(psrs, Nvecs, Ts, device="cpu")
+    eng.precompute(np.arange(1, 41) / pta.Tspan)
+    eng.enable_draw_compression(
+        [c.var_slice for c in pta.rn_containers],
+        [c.get_phiinv(noise) for c in pta.rn_containers],
+    )
+    assert all(blk.comp is not None for blk in eng.blocks), \
+        "probe must keep compression for the healthy benchmark model"
