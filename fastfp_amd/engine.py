@@ -228,18 +228,25 @@ class FpEngine:
         variable bins are ignored/zeroed).
         """
         assert self.freqs is not None, "call precompute(freqs) first"
+        # The K/M0/N0/G setup runs on CPU LAPACK even for GPU engines:
+        # rocBLAS trsm is inversion-based rather than backward-stable
+        # substitution and measured ~50x less accurate here (1e-4 vs
+        # 2e-6 of spectrum scale at the benchmark conditioning) — a
+        # one-time ~2 MB/pulsar round trip buys LAPACK accuracy.
+        wdev = torch.device("cpu")
         for blk, sl, pf in zip(self.blocks, var_slices, phiinv_fixed):
             m = blk.m
-            pf = _t64(pf, self.device).clone()
+            pf = _t64(pf, wdev).clone()
             # Sigma_0 must be SPD even when timing-model columns are
             # (nearly) degenerate with red-noise Fourier columns, so the
             # variable bins keep a TINY reference prior delta_i
             # proportional to the Sigma diagonal; the per-draw
             # correction then uses Delta_d = phiinv_d - delta, valid
             # while phiinv_d >> delta (checked by compression_margin).
-            delta0 = jitter_rel * torch.diagonal(blk.TNT)[sl].abs()
+            TNTc = blk.TNT.to(wdev)
+            delta0 = jitter_rel * torch.diagonal(TNTc)[sl].abs()
             pf[sl] = delta0
-            sigma0 = blk.TNT + torch.diag(pf)
+            sigma0 = TNTc + torch.diag(pf)
             L0 = torch.linalg.cholesky(sigma0)
             # rank-deficient TNT (basis larger than the TOA count)
             # leaves Sigma_0 supported only by the jitter along its
@@ -250,15 +257,16 @@ class FpEngine:
             if blk.ntoa < m:
                 blk.comp = None
                 continue
-            RHSe = blk.RHS[:m, :]  # (m, 2F+1)
+            RHSe = blk.RHS[:m, :].to(wdev)  # (m, 2F+1)
             ncols = RHSe.shape[1]
             mv = len(range(*sl.indices(m)))
-            # column-chunked solves: rocBLAS trsm wants per-call
-            # workspace ~ n, which fails outright at the 2e5-frequency
-            # SKA shape
-            K = torch.empty((mv, ncols), dtype=torch.float64, device=self.device)
-            M0 = torch.empty((3, ncols // 2), dtype=torch.float64, device=self.device)
-            N0 = torch.empty((2, ncols // 2), dtype=torch.float64, device=self.device)
+            sNs_c = blk.sNs.to(wdev)
+            sNr_c = blk.sNr.to(wdev)
+            # column-chunked solves (bounds LAPACK workspace/temporaries
+            # at the 2e5-frequency SKA shape)
+            K = torch.empty((mv, ncols), dtype=torch.float64)
+            M0 = torch.empty((3, ncols // 2), dtype=torch.float64)
+            N0 = torch.empty((2, ncols // 2), dtype=torch.float64)
             CH = 16384  # even: chunks stay aligned to sin/cos pairs
             # first pass: the u column (last), needed by every chunk
             wcol = torch.linalg.solve_triangular(
@@ -276,31 +284,31 @@ class FpEngine:
                 Ws = W0c[:, 0 : 2 * (fe - fs) : 2]
                 Wc = W0c[:, 1 : 2 * (fe - fs) : 2]
                 wu = wcol[:, 0]
-                M0[0, fs:fe] = blk.sNs[0, fs:fe] - (Ws * Ws).sum(0)
-                M0[1, fs:fe] = blk.sNs[1, fs:fe] - (Wc * Wc).sum(0)
-                M0[2, fs:fe] = blk.sNs[2, fs:fe] - (Ws * Wc).sum(0)
-                N0[0, fs:fe] = blk.sNr[0, fs:fe] - Ws.T @ wu
-                N0[1, fs:fe] = blk.sNr[1, fs:fe] - Wc.T @ wu
+                M0[0, fs:fe] = sNs_c[0, fs:fe] - (Ws * Ws).sum(0)
+                M0[1, fs:fe] = sNs_c[1, fs:fe] - (Wc * Wc).sum(0)
+                M0[2, fs:fe] = sNs_c[2, fs:fe] - (Ws * Wc).sum(0)
+                N0[0, fs:fe] = sNr_c[0, fs:fe] - Ws.T @ wu
+                N0[1, fs:fe] = sNr_c[1, fs:fe] - Wc.T @ wu
                 S0c = torch.linalg.solve_triangular(
                     L0.transpose(0, 1), W0c, upper=True
                 )
                 K[:, lo:hi] = S0c[sl, :]
-            M0 = M0.contiguous()
-            N0 = N0.contiguous()
             S0inv = torch.cholesky_inverse(L0)
-            G = S0inv[sl, sl].contiguous()
+            G = S0inv[sl, sl]
             if self._use_hip:
                 from fastfp_amd import ops
 
                 mvp = ops.check_m(mv)
-                Kp = torch.zeros(
-                    (mvp, K.shape[1]), dtype=torch.float64, device=self.device
-                )
+                Kp = torch.zeros((mvp, K.shape[1]), dtype=torch.float64)
                 Kp[:mv] = K
                 K = Kp
             blk.comp = dict(
-                G=G, K=K.contiguous(), M0=M0, N0=N0, var=sl, mv=mv,
-                delta0=delta0,
+                G=G.to(self.device).contiguous(),
+                K=K.to(self.device).contiguous(),
+                M0=M0.to(self.device).contiguous(),
+                N0=N0.to(self.device).contiguous(),
+                var=sl, mv=mv,
+                delta0=delta0.to(self.device),
             )
         self._probe_compression(phiinv_fixed)
         self._stack_compression()
