@@ -413,3 +413,35 @@ def test_graph_captured_sweep_bitwise_equals_eager():
     g.replay()
     torch.cuda.synchronize()
     np.testing.assert_array_equal(fp_accum.cpu().numpy(), eager)
+
+
+def test_probe_keeps_compression_at_bench_conditioning():
+    """GPU regression guard for the compression-precompute accuracy:
+    at the benchmark's conditioning (rn 30 + gwb 14 components, dense
+    TOA coverage) the enable-time probe must keep EVERY pulsar on the
+    compressed path.  This failed when the K/M0/N0/G setup ran through
+    rocBLAS solve_triangular (inversion-based, ~1e-4 spectrum-scale
+    error) instead of CPU LAPACK — docs/TUNING_NOTES.md."""
+    from fastfp_amd import FpEngine, get_mats_nmfp, initialize_pta, \
+        make_synthetic_pta
+
+    psrs = make_synthetic_pta(npsr=6, ntoa=3000, tspan_yr=15.0, ntm=8,
+                              seed=1234, ragged=True)
+    noise = {"gw_gamma": 13.0 / 3.0, "gw_log10_A": float(np.log10(2e-15))}
+    for p in psrs:
+        noise[f"{p.name}_red_noise_gamma"] = 13.0 / 3.0
+        noise[f"{p.name}_red_noise_log10_A"] = -14.5
+    pta = initialize_pta(psrs, noise, inc_cp=True, rn_comps=30,
+                         gwb_comps=14)
+    TNTs, Nvecs, Ts = get_mats_nmfp(pta, noise)
+    eng = FpEngine(psrs, Nvecs, Ts, device=DEV)
+    eng.precompute(np.arange(1, 201) / pta.Tspan)
+    probe = {k: v for k, v in noise.items() if k in pta.params}
+    eng.enable_draw_compression(
+        [c.var_slice for c in pta.rn_containers],
+        [c.get_phiinv(probe).to(DEV) for c in pta.rn_containers],
+    )
+    kept = sum(blk.comp is not None for blk in eng.blocks)
+    assert kept == len(eng.blocks), \
+        f"probe disabled compression on {len(eng.blocks) - kept} pulsars"
+    assert eng._comp_stack is not None, "stacked compressed path must form"
