@@ -355,3 +355,49 @@ def test_probe_keeps_compression_on_healthy_model():
     )
     assert all(blk.comp is not None for blk in eng.blocks), \
         "probe must keep compression for the healthy benchmark model"
+
+
+def test_mixed_compression_rank_deficient_pulsar():
+    """A pulsar with fewer TOAs than basis columns must stay on the
+    exact direct path (Sigma_0 is jitter-supported along the null
+    space) while the rest keep the compressed path — and the MIXED
+    sweep must match the fully-direct sweep."""
+    import torch
+
+    from fastfp_amd.noise import batch_phiinv, check_batch_homogeneous
+
+    psrs = make_synthetic_pta(npsr=2, ntoa=90, ntm=4, seed=23) + \
+        make_synthetic_pta(npsr=1, ntoa=10, ntm=4, seed=24, ragged=False)
+    noise = {"gw_gamma": 13.0 / 3.0, "gw_log10_A": float(np.log10(2e-15))}
+    for p in psrs:
+        noise[f"{p.name}_red_noise_gamma"] = 4.0
+        noise[f"{p.name}_red_noise_log10_A"] = -14.5
+    pta = initialize_pta(psrs, noise, inc_cp=True, rn_comps=5, gwb_comps=4)
+    TNTs, Nvecs, Ts = get_mats_nmfp(pta, noise)
+    eng = FpEngine(psrs, Nvecs, Ts, device="cpu")
+    freqs = np.linspace(4e-9, 5e-8, 6)
+    eng.precompute(freqs)
+    eng.enable_draw_compression(
+        [c.var_slice for c in pta.rn_containers],
+        [c.get_phiinv(noise) for c in pta.rn_containers],
+    )
+    kept = [blk.comp is not None for blk in eng.blocks]
+    assert kept == [True, True, False], kept
+
+    D = 4
+    rng = np.random.default_rng(31)
+    pool = {
+        n: torch.as_tensor(
+            rng.uniform(2, 6, D) if n.endswith("gamma")
+            else rng.uniform(-16, -14, D), dtype=torch.float64)
+        for n in pta.params
+    }
+    homog = check_batch_homogeneous(pta.rn_containers)
+    phiinvs = batch_phiinv(pta.rn_containers, pool, homogeneous=homog)
+    mixed = torch.zeros((D, 6), dtype=torch.float64)
+    eng.sweep(phiinvs=phiinvs, draw_chunk=4, accumulate_to=mixed)
+
+    eng.disable_draw_compression()
+    direct = torch.zeros_like(mixed)
+    eng.sweep(phiinvs=phiinvs, draw_chunk=4, accumulate_to=direct)
+    np.testing.assert_allclose(mixed.numpy(), direct.numpy(), rtol=1e-7)
