@@ -401,3 +401,26 @@ def test_mixed_compression_rank_deficient_pulsar():
     direct = torch.zeros_like(mixed)
     eng.sweep(phiinvs=phiinvs, draw_chunk=4, accumulate_to=direct)
     np.testing.assert_allclose(mixed.numpy(), direct.numpy(), rtol=1e-7)
+
+
+def test_precompute_invalidates_stale_compression():
+    """Re-running precompute on a NEW frequency grid must drop
+    compression state built against the old grid (K's columns are
+    per-frequency)."""
+    psrs = make_synthetic_pta(npsr=2, ntoa=60, ntm=3, seed=40)
+    noise = {"gw_gamma": 13.0 / 3.0, "gw_log10_A": float(np.log10(2e-15))}
+    for p in psrs:
+        noise[f"{p.name}_red_noise_gamma"] = 4.0
+        noise[f"{p.name}_red_noise_log10_A"] = -14.5
+    pta = initialize_pta(psrs, noise, inc_cp=True, rn_comps=3, gwb_comps=2)
+    TNTs, Nvecs, Ts = get_mats_nmfp(pta, noise)
+    eng = FpEngine(psrs, Nvecs, Ts, device="cpu")
+    eng.precompute(np.linspace(4e-9, 5e-8, 4))
+    eng.enable_draw_compression(
+        [c.var_slice for c in pta.rn_containers],
+        [c.get_phiinv(noise) for c in pta.rn_containers],
+    )
+    assert all(blk.comp is not None for blk in eng.blocks)
+    eng.precompute(np.linspace(4e-9, 5e-8, 9))
+    assert all(blk.comp is None for blk in eng.blocks)
+    assert eng._comp_stack is None
