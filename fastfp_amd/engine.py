@@ -243,19 +243,24 @@ class FpEngine:
             # proportional to the Sigma diagonal; the per-draw
             # correction then uses Delta_d = phiinv_d - delta, valid
             # while phiinv_d >> delta (checked by compression_margin).
+            # rank-deficient TNT (basis larger than the TOA count)
+            # leaves Sigma_0 supported only by the jitter along its
+            # null space — keep such pulsars on the exact direct path
+            # (checked BEFORE factoring: their Sigma_0 may not even be
+            # numerically PD).  Near-degeneracy SHORT of deficiency is
+            # handled by the empirical probe below: Cholesky-pivot
+            # heuristics measured uncorrelated with the actual error.
+            if blk.ntoa < m:
+                blk.comp = None
+                continue
             TNTc = blk.TNT.to(wdev)
             delta0 = jitter_rel * torch.diagonal(TNTc)[sl].abs()
             pf[sl] = delta0
             sigma0 = TNTc + torch.diag(pf)
-            L0 = torch.linalg.cholesky(sigma0)
-            # rank-deficient TNT (basis larger than the TOA count)
-            # leaves Sigma_0 supported only by the jitter along its
-            # null space — keep such pulsars on the exact direct path.
-            # (Near-degeneracy SHORT of deficiency is handled by the
-            # empirical probe below: Cholesky-pivot heuristics measured
-            # uncorrelated with the actual compression error.)
-            if blk.ntoa < m:
-                blk.comp = None
+            try:
+                L0 = torch.linalg.cholesky(sigma0)
+            except torch.linalg.LinAlgError:
+                blk.comp = None  # fail safe to the exact direct path
                 continue
             RHSe = blk.RHS[:m, :].to(wdev)  # (m, 2F+1)
             ncols = RHSe.shape[1]
