@@ -83,6 +83,7 @@ def main(
     seed=0,
     resume=False,
     ecorr_kernel=False,
+    checkpoint=True,
 ):
     logging.basicConfig(format="%(levelname)s: %(message)s", level=logging.INFO)
     logger = logging.getLogger(__name__)
@@ -142,9 +143,55 @@ def main(
     my_idx = np.arange(nsamples)[my]
 
     batch_dir = os.path.join(outdir, f".{savefile}.batches")
+    # Resume safety: batch shards are keyed by (rank, batch offset)
+    # only, so a manifest fingerprints everything that changes their
+    # CONTENT — re-running with a different seed/chain/model silently
+    # mixing stale shards into the output would be far worse than
+    # recomputing.  On mismatch the stale shards are ignored (resume
+    # effectively restarts) and the manifest is rewritten.
+    import hashlib
+
+    try:
+        chain_stat = os.stat(chainfile)
+        chain_fp = f"{chain_stat.st_size}:{chain_stat.st_mtime_ns}"
+    except OSError:
+        chain_fp = "unknown"
+    manifest_key = hashlib.sha256(
+        repr(
+            dict(
+                seed=seed, nsamples=nsamples, batch_size=batch_size,
+                world=world, ncwfreqs=ncwfreqs, chain=chain_fp,
+                inc_ecorr=inc_ecorr, inc_cp=inc_cp, nrncomps=nrncomps,
+                ngwbcomps=ngwbcomps, ecorr_kernel=ecorr_kernel,
+                psrfile=os.path.abspath(psrfile),
+                noisefile=os.path.abspath(noisefile),
+            )
+        ).encode()
+    ).hexdigest()[:16]
+    manifest_path = os.path.join(batch_dir, "MANIFEST")
     if rank == 0:
         os.makedirs(outdir, exist_ok=True)
+    if rank == 0 and checkpoint:
         os.makedirs(batch_dir, exist_ok=True)
+        prev = None
+        if checkpoint and os.path.exists(manifest_path):
+            with open(manifest_path) as f:
+                prev = f.read().strip()
+        if checkpoint and prev != manifest_key:
+            stale = [
+                fn for fn in os.listdir(batch_dir)
+                if fn.endswith(".npy")
+            ]
+            if stale and prev is not None:
+                logger.warning(
+                    "resume: run configuration changed (manifest %s -> %s); "
+                    "removing %d stale batch shards",
+                    prev, manifest_key, len(stale),
+                )
+            for fn in stale:
+                os.remove(os.path.join(batch_dir, fn))
+            with open(manifest_path, "w") as f:
+                f.write(manifest_key)
     if world > 1:
         torch.distributed.barrier()
 
@@ -168,16 +215,28 @@ def main(
                        disable=rank != 0, desc="draw batches")
     except ImportError:
         batches = range(0, len(my_idx), batch_size)
+    # checkpoint writes happen on a background thread so disk I/O
+    # overlaps the next batch's GPU sweep (the CLI-vs-bench gap was
+    # dominated by synchronous np.save + host gather; docs/PERFORMANCE.md)
+    from concurrent.futures import ThreadPoolExecutor
+
+    writer = ThreadPoolExecutor(max_workers=1) if checkpoint else None
+    pending = []
     for lo in batches:
         sel = my_idx[lo : lo + batch_size]
         ck = os.path.join(batch_dir, f"r{rank}_b{lo}.npy")
-        if resume and os.path.exists(ck):
+        if checkpoint and resume and os.path.exists(ck):
             parts.append(np.load(ck))
             continue
         samples = map_params(pta, rns_full[:, sel])
         vals = nmfp.sweep(freqs, samples, Nvecs, Ts, engine=eng)
-        np.save(ck, vals)
+        if writer is not None:
+            pending.append(writer.submit(np.save, ck, vals))
         parts.append(vals)
+    if writer is not None:
+        for fut in pending:
+            fut.result()  # surface write errors before declaring success
+        writer.shutdown()
     local_vals = (
         np.vstack(parts) if parts else np.zeros((0, ncwfreqs))
     )
@@ -222,6 +281,10 @@ def cli():
                         help="model ECORR as block-diagonal white noise "
                              "(EcorrKernelNoise; the reference's "
                              "unsupported case)")
+    parser.add_argument("--no-checkpoint", dest="checkpoint",
+                        action="store_false",
+                        help="disable per-batch checkpoint shards "
+                             "(faster; --resume unavailable)")
     main(**vars(parser.parse_args()))
 
 

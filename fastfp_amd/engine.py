@@ -153,7 +153,7 @@ class FpEngine:
         with _roctx("fastfp:freq_precompute"):
             for blk in self.blocks:
                 if self._use_hip:
-                    self._precompute_hip(blk, freqs, freq_chunk)
+                    self._precompute_hip(blk, freqs)
                 else:
                     self._precompute_eager(blk, freqs, freq_chunk)
         self._stack_direct()
@@ -189,12 +189,14 @@ class FpEngine:
         RHS[:, -1] = blk.TNr
         blk.RHS, blk.sNs, blk.sNr = RHS, sNs, sNr
 
-    def _precompute_hip(self, blk: PulsarBlock, freqs, freq_chunk):
+    def _precompute_hip(self, blk: PulsarBlock, freqs):
         from fastfp_amd import ops
 
         if blk.block_noise is None:
+            # no freq_chunk: the HIP kernels generate the S panel in
+            # LDS and need no chunking (ops.freq_precompute docstring)
             blk.RHS, blk.sNs, blk.sNr = ops.freq_precompute(
-                blk.toas, blk.Nvec, blk.r, blk.T, blk.TNr, freqs, freq_chunk
+                blk.toas, blk.Nvec, blk.r, blk.T, blk.TNr, freqs
             )
         else:
             blk.RHS, blk.sNs, blk.sNr = ops.freq_precompute_block(

@@ -52,11 +52,24 @@ def white_noise_nvec(psr, noise: dict = None, simple_wn: bool = True,
     ``simple_wn``: EFAC = 1.0 (the reference's simulated-data default,
     ``/root/reference/fastfp/utils.py:147-149``): N = toaerr^2.
 
-    Otherwise per-backend EFAC/EQUAD with the TEMPO2 (t2equad)
-    convention used by enterprise's ``white_noise_block``:
-    ``N = efac^2 * (toaerr^2 + t2equad^2)`` with parameters read from the
-    noise dict as ``{psr}_{backend}_efac`` / ``{psr}_{backend}_log10_t2equad``
-    (missing keys default to efac=1, equad=0).
+    Otherwise per-backend EFAC/EQUAD with all three EQUAD conventions
+    enterprise's ``white_noise_block`` admits
+    (``/root/reference/fastfp/utils.py:151-155``), keyed per backend
+    group ``{psr}_{backend}`` (or ``{psr}`` when ``select != "backend"``):
+
+    - TEMPO2 convention (``_log10_t2equad``):
+      ``N = efac^2 * (toaerr^2 + t2equad^2)``
+    - TempoNest convention (``_log10_tnequad``):
+      ``N = (efac * toaerr)^2 + tnequad^2``
+    - legacy key (``_log10_equad``): historical NANOGrav noise dicts;
+      enterprise's old ``EquadNoise`` added it in quadrature AFTER the
+      EFAC scaling, i.e. the TempoNest convention:
+      ``N = (efac * toaerr)^2 + equad^2``.
+
+    Precedence when several keys are present: t2equad > tnequad >
+    legacy equad (enterprise's white_noise_block instantiates exactly
+    one of these per model, so overlapping keys indicate a mixed dict —
+    the modern key wins).  Missing keys default to efac=1, equad=0.
     """
     if simple_wn:
         return psr.toaerrs**2
@@ -73,9 +86,16 @@ def white_noise_nvec(psr, noise: dict = None, simple_wn: bool = True,
         groups = [(np.ones(psr.ntoa, dtype=bool), psr.name)]
     for mask, prefix in groups:
         efac = float(noise.get(f"{prefix}_efac", 1.0))
-        l10eq = noise.get(f"{prefix}_log10_t2equad", None)
-        equad2 = 10.0 ** (2.0 * float(l10eq)) if l10eq is not None else 0.0
-        nvec[mask] = efac**2 * (psr.toaerrs[mask] ** 2 + equad2)
+        t2 = noise.get(f"{prefix}_log10_t2equad", None)
+        tn = noise.get(f"{prefix}_log10_tnequad", None)
+        if tn is None:
+            tn = noise.get(f"{prefix}_log10_equad", None)  # legacy key
+        if t2 is not None:
+            equad2 = 10.0 ** (2.0 * float(t2))
+            nvec[mask] = efac**2 * (psr.toaerrs[mask] ** 2 + equad2)
+        else:
+            equad2 = 10.0 ** (2.0 * float(tn)) if tn is not None else 0.0
+            nvec[mask] = (efac * psr.toaerrs[mask]) ** 2 + equad2
     return nvec
 
 

@@ -75,7 +75,7 @@ def check_m(m: int) -> int:
 # ----------------------------------------------------------------------
 # op wrappers (thin; shapes documented in the kernels)
 # ----------------------------------------------------------------------
-def freq_precompute(toas, Nvec, r, T, TNr, freqs, freq_chunk: int = 8192):
+def freq_precompute(toas, Nvec, r, T, TNr, freqs):
     """GPU frequency precompute for one pulsar.
 
     Returns (RHS (mp, 2F+1), sNs (3, F), sNr (2, F)).  The fused
@@ -83,6 +83,10 @@ def freq_precompute(toas, Nvec, r, T, TNr, freqs, freq_chunk: int = 8192):
     fused signal-basis MFMA DGEMM (sbgemm) computes B = T^T (N^-1 S)
     with the S panel generated in LDS — NS is never materialized.
     RHS rows are zero-padded to mp (multiple of 16).
+
+    Unlike the eager path there is no ``freq_chunk`` knob: the sin/cos
+    S panel is generated tile-by-tile in LDS inside the kernels and no
+    (F, ntoa) intermediate ever exists to bound.
     """
     ext = _try_load()
     F = int(freqs.shape[0])

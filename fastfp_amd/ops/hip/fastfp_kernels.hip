@@ -110,7 +110,7 @@ extern "C" __global__ __launch_bounds__(256) void sigdots_kernel(
 // the T-panel is LDS-staged.  Output is the (mp x 2F) RHS block,
 // written directly (ksplit==1, ld = ldo) or into per-split partial
 // planes (deterministic torch reduction afterwards).
-// grid.x = ceil(2F / 64) col tiles, grid.y = ksplit;  block = 256 (4 waves)
+// grid.x = ceil(2F / 64) col tiles, grid.y = ksplit;  block = 512 (8 waves)
 // ---------------------------------------------------------------------
 extern "C" __global__ __launch_bounds__(512) void sbgemm_kernel(
     const double* __restrict__ T /*(ntoa, m) row-major*/,

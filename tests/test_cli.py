@@ -77,3 +77,44 @@ def test_run_nmfp_resume(inputs):
     run_nmfp.main(psrfile, noisefile, chainfile, "nm", resume=True, **kwargs)
     b = np.load(os.path.join(outdir, "nm.npy"))
     np.testing.assert_array_equal(a, b)
+
+
+def test_run_nmfp_resume_invalidates_on_config_change(inputs):
+    """A resume with a changed seed must NOT mix stale batch shards into
+    the output (the shards are keyed only by rank/offset; the manifest
+    fingerprints the run config)."""
+    tmp, psrfile, noisefile, chainfile = inputs
+    outdir = str(tmp / "res3")
+    kwargs = dict(
+        inc_cp=False, nrncomps=3, ncwfreqs=3, nsamples=4, batch_size=2,
+        outdir=outdir, device="cpu",
+    )
+    run_nmfp.main(psrfile, noisefile, chainfile, "nm", seed=3, **kwargs)
+    # different seed -> different draw selection -> shards are stale
+    run_nmfp.main(psrfile, noisefile, chainfile, "nm", seed=4, resume=True,
+                  **kwargs)
+    b = np.load(os.path.join(outdir, "nm.npy"))
+    # fresh non-resumed run with seed=4 is the ground truth
+    outdir2 = str(tmp / "res3b")
+    kwargs["outdir"] = outdir2
+    run_nmfp.main(psrfile, noisefile, chainfile, "nm", seed=4, **kwargs)
+    c = np.load(os.path.join(outdir2, "nm.npy"))
+    np.testing.assert_array_equal(b, c)
+
+
+def test_run_nmfp_no_checkpoint(inputs):
+    """--no-checkpoint skips shard writes but produces identical output."""
+    tmp, psrfile, noisefile, chainfile = inputs
+    outdir = str(tmp / "res4")
+    kwargs = dict(
+        inc_cp=False, nrncomps=3, ncwfreqs=3, nsamples=4, batch_size=2,
+        device="cpu", seed=5,
+    )
+    run_nmfp.main(psrfile, noisefile, chainfile, "nm", outdir=outdir, **kwargs)
+    a = np.load(os.path.join(outdir, "nm.npy"))
+    outdir2 = str(tmp / "res4b")
+    run_nmfp.main(psrfile, noisefile, chainfile, "nm", outdir=outdir2,
+                  checkpoint=False, **kwargs)
+    b = np.load(os.path.join(outdir2, "nm.npy"))
+    np.testing.assert_array_equal(a, b)
+    assert not os.path.exists(os.path.join(outdir2, ".nm.batches"))

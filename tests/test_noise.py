@@ -141,6 +141,63 @@ def test_white_noise_nvec_backends(psr):
     np.testing.assert_allclose(full, want, rtol=1e-12)
 
 
+def test_white_noise_nvec_tnequad_and_legacy(psr):
+    """TempoNest (tnequad) and legacy (log10_equad) conventions on a
+    real-shaped NANOGrav noise dict — enterprise's white_noise_block
+    admits all three (/root/reference/fastfp/utils.py:151-155).  The
+    legacy key uses tnequad semantics (old EquadNoise: quadrature AFTER
+    the EFAC scaling)."""
+    backends = np.unique(psr.backend_flags)
+    # tnequad convention: N = (efac*sig)^2 + tnequad^2
+    noise = {}
+    for b in backends:
+        noise[f"{psr.name}_{b}_efac"] = 1.3
+        noise[f"{psr.name}_{b}_log10_tnequad"] = -6.2
+    nv = white_noise_nvec(psr, noise, simple_wn=False)
+    want = (1.3 * psr.toaerrs) ** 2 + 10.0 ** (2 * -6.2)
+    np.testing.assert_allclose(nv, want, rtol=1e-12)
+
+    # legacy log10_equad keys: same tnequad semantics
+    noise = {}
+    for b in backends:
+        noise[f"{psr.name}_{b}_efac"] = 1.1
+        noise[f"{psr.name}_{b}_log10_equad"] = -6.8
+    nv = white_noise_nvec(psr, noise, simple_wn=False)
+    want = (1.1 * psr.toaerrs) ** 2 + 10.0 ** (2 * -6.8)
+    np.testing.assert_allclose(nv, want, rtol=1e-12)
+
+
+def test_white_noise_nvec_key_precedence(psr):
+    """Mixed dicts: t2equad wins over tnequad wins over legacy equad;
+    conventions may differ per backend group."""
+    backends = np.unique(psr.backend_flags)
+    assert len(backends) >= 2
+    b0, b1 = backends[0], backends[1]
+    noise = {
+        # b0 carries BOTH t2equad and legacy equad -> t2equad wins
+        f"{psr.name}_{b0}_efac": 1.4,
+        f"{psr.name}_{b0}_log10_t2equad": -6.0,
+        f"{psr.name}_{b0}_log10_equad": -5.0,
+        # b1 carries tnequad and legacy equad -> tnequad wins
+        f"{psr.name}_{b1}_efac": 0.9,
+        f"{psr.name}_{b1}_log10_tnequad": -6.4,
+        f"{psr.name}_{b1}_log10_equad": -5.5,
+    }
+    nv = white_noise_nvec(psr, noise, simple_wn=False)
+    m0 = psr.backend_flags == b0
+    m1 = psr.backend_flags == b1
+    np.testing.assert_allclose(
+        nv[m0], 1.4**2 * (psr.toaerrs[m0] ** 2 + 1e-12), rtol=1e-12
+    )
+    np.testing.assert_allclose(
+        nv[m1], (0.9 * psr.toaerrs[m1]) ** 2 + 10.0 ** (2 * -6.4), rtol=1e-12
+    )
+    # untouched backends stay at efac=1, equad=0
+    rest = ~(m0 | m1)
+    if rest.any():
+        np.testing.assert_allclose(nv[rest], psr.toaerrs[rest] ** 2, rtol=1e-12)
+
+
 def test_white_noise_no_selection(psr):
     """select != backend: one parameter set per pulsar with the
     un-selected enterprise naming ({psr}_efac)."""
