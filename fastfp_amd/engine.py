@@ -331,15 +331,33 @@ class FpEngine:
         for blk, pf in zip(self.blocks, phiinv_fixed):
             if blk.comp is None:
                 continue
-            if blk.m > 128 and self._use_hip:
-                continue  # no direct GPU reference; compression required
             pinv = _t64(pf, self.device).reshape(1, -1)
             c = blk.comp
             fpA = torch.zeros((1, F), dtype=torch.float64, device=self.device)
             fpB = torch.zeros_like(fpA)
             phi_var = (1.0 / (pinv[:, c["var"]] - c["delta0"][None, :])).contiguous()
             try:
-                if self._use_hip:
+                if self._use_hip and blk.m > 128:
+                    # no direct GPU kernel above m=128: the probe's
+                    # direct reference runs through the CPU LAPACK eager
+                    # path on this one draw (compression precompute is
+                    # already pinned to CPU, so this adds one more small
+                    # host round trip at setup time only)
+                    from fastfp_amd import ops
+
+                    ops.chol_trsm_fp_accum(
+                        c["G"], phi_var, c["K"], c["M0"], c["N0"], fpA,
+                        gsign=-1.0,
+                    )
+                    pinv_c = pinv.cpu()
+                    sigma_c = blk.TNT.cpu()[None, :, :] + torch.diag_embed(pinv_c)
+                    fpB_c = torch.zeros((1, F), dtype=torch.float64)
+                    self._accum_eager_mats(
+                        sigma_c, blk.RHS[: blk.m].cpu(), blk.sNs.cpu(),
+                        blk.sNr.cpu(), fpB_c, 1.0,
+                    )
+                    fpB = fpB_c.to(self.device)
+                elif self._use_hip:
                     from fastfp_amd import ops
 
                     ops.chol_trsm_fp_accum(

@@ -95,7 +95,26 @@ class FastFp:
             eng.enable_draw_compression(
                 [c.var_slice for c in conts], phiinvs
             )
-            fp = eng.sweep(phiinvs=phiinvs)
+            # the compressed route is the ONLY m>128 GPU path, so guard
+            # it twice: the accuracy probe may have dropped pulsars
+            # (blk.comp None -> their direct solve would exceed the
+            # kernel cap), and near-degenerate priors shrink the
+            # phi/jitter margin the Woodbury correction relies on.  In
+            # either case fall back to the CPU LAPACK engine — slower
+            # but always exact.
+            margin = eng.compression_margin(phiinvs)
+            if any(blk.comp is None for blk in eng.blocks) or margin < 1e3:
+                import warnings
+
+                warnings.warn(
+                    "m > 128 GPU sweep: compression unsafe "
+                    f"(margin={margin:.1e}); falling back to the CPU engine"
+                )
+                eng = FpEngine(self.psrs, Nvecs, Ts, device="cpu")
+                eng.precompute(freqs, freq_chunk=freq_chunk)
+                fp = eng.sweep(sigmas=sigmas)
+            else:
+                fp = eng.sweep(phiinvs=phiinvs)
         else:
             fp = eng.sweep(sigmas=sigmas)
         return fp.cpu().numpy()

@@ -26,8 +26,13 @@ def init_distributed(backend: str = None, device: torch.device = None):
     a single-process run: returns (0, 1, device) without init.
     Backend default: "nccl" (= RCCL on ROCm) when CUDA devices are
     visible, else "gloo".
+
+    A torchrun launch with WORLD_SIZE=1 still initializes the process
+    group: the RCCL communicator setup and every collective call site
+    then run exactly as in the multi-GPU case (one-rank collectives are
+    cheap), so a 1-GPU box exercises the full distributed path.
     """
-    if "RANK" not in os.environ or int(os.environ.get("WORLD_SIZE", "1")) == 1:
+    if "RANK" not in os.environ:
         if device is None:
             device = torch.device("cuda:0" if torch.cuda.is_available() else "cpu")
         return 0, 1, device
@@ -67,9 +72,11 @@ def all_gather_concat(local: torch.Tensor, world: int, dim: int = 0) -> torch.Te
     """All-gather variable-length shards along ``dim`` and concatenate.
 
     Uses all_gather with per-rank padded buffers (shard sizes may differ
-    by 1 from block sharding).
+    by 1 from block sharding).  With an initialized 1-rank group the
+    collectives still run (exercising the RCCL path); without a group
+    this is the identity.
     """
-    if world == 1 or not dist.is_initialized():
+    if not dist.is_initialized():
         return local
     local = local.contiguous()
     n_local = torch.tensor([local.shape[dim]], dtype=torch.int64, device=local.device)

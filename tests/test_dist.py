@@ -5,11 +5,12 @@ sharded + all-gathered spectrum must equal the single-process result
 import os
 
 import numpy as np
+import pytest
 import torch
 import torch.multiprocessing as mp
 
 from fastfp_amd import FpEngine, get_mats_fp, initialize_pta, make_synthetic_pta
-from fastfp_amd.parallel import all_gather_concat, shard_slice
+from fastfp_amd.parallel import all_gather_concat, init_distributed, shard_slice
 
 
 def _build():
@@ -52,6 +53,69 @@ def test_freq_sharded_fp_matches_single(tmp_path):
     # identical partition of identical per-frequency computations:
     # bitwise equality expected (fp64, per-frequency independence)
     np.testing.assert_array_equal(got, want)
+
+
+def _ws1_env(port):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    os.environ["RANK"] = "0"
+    os.environ["WORLD_SIZE"] = "1"
+    os.environ["LOCAL_RANK"] = "0"
+
+
+def _ws1_env_clear():
+    for k in ("MASTER_ADDR", "MASTER_PORT", "RANK", "WORLD_SIZE", "LOCAL_RANK"):
+        os.environ.pop(k, None)
+
+
+def test_ws1_initialized_group_collectives_cpu():
+    """A torchrun-style WORLD_SIZE=1 launch now initializes the process
+    group, and every collective call site runs (one-rank collectives):
+    the gathered spectrum must equal the local one bitwise."""
+    _ws1_env(29861)
+    try:
+        rank, world, dev = init_distributed(device=torch.device("cpu"))
+        assert (rank, world) == (0, 1)
+        assert torch.distributed.is_initialized()
+        x = torch.arange(12, dtype=torch.float64).reshape(3, 4)
+        got = all_gather_concat(x.clone(), world, dim=0)
+        np.testing.assert_array_equal(got.numpy(), x.numpy())
+        t = torch.tensor([3.5], dtype=torch.float64)
+        torch.distributed.all_reduce(t, op=torch.distributed.ReduceOp.MAX)
+        assert float(t.item()) == 3.5
+        torch.distributed.barrier()
+    finally:
+        if torch.distributed.is_initialized():
+            torch.distributed.destroy_process_group()
+        _ws1_env_clear()
+
+
+@pytest.mark.gpu
+def test_ws1_rccl_collectives_on_device():
+    """RCCL (nccl backend on ROCm) communicator init + collectives on
+    HIP tensors at world_size 1 — exercises the exact code path the
+    8-GPU bench uses (init_distributed -> all_gather_concat ->
+    all_reduce MAX on device tensors) on a 1-GPU box."""
+    if not torch.cuda.is_available():
+        pytest.skip("needs a GPU")
+    _ws1_env(29862)
+    try:
+        rank, world, dev = init_distributed()
+        assert dev.type == "cuda"
+        assert torch.distributed.get_backend() in ("nccl", "cclx")
+        x = torch.randn(512, 7, dtype=torch.float64, device=dev)
+        got = all_gather_concat(x.clone(), world, dim=0)
+        torch.cuda.synchronize()
+        np.testing.assert_array_equal(got.cpu().numpy(), x.cpu().numpy())
+        t = torch.tensor([2.25], dtype=torch.float64, device=dev)
+        torch.distributed.all_reduce(t, op=torch.distributed.ReduceOp.MAX)
+        torch.distributed.barrier()
+        torch.cuda.synchronize()
+        assert float(t.item()) == 2.25
+    finally:
+        if torch.distributed.is_initialized():
+            torch.distributed.destroy_process_group()
+        _ws1_env_clear()
 
 
 def test_shard_slice_partitions():
