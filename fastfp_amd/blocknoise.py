@@ -5,10 +5,25 @@ BASELINE.json config 4).
 
 Per observing epoch (and backend), ECORR adds a fully-correlated
 component to the white noise: the per-epoch block is
-``N_b = diag(nvec_b) + 10^(2*log10_ecorr) * J`` (J = all-ones).  This
-module partitions the TOAs into such blocks, permutes them contiguous,
-factors each block (Cholesky) and precomputes the dense block inverses,
-and applies ``N^{-1}`` to vectors/panels.
+``N_b = diag(nvec_b) + 10^(2*log10_ecorr) * J`` (J = all-ones) —
+**diagonal plus rank-1**, so Sherman–Morrison gives the exact inverse
+in closed form with O(n) work and no per-block factorization:
+
+    N_b^{-1} = D^{-1} - beta_b * d d^T,   d = D^{-1} 1,
+    beta_b   = e2 / (1 + e2 * s_b),       s_b = sum_i 1/nvec_i,
+    logdet   = sum_i log(nvec_i) + log1p(e2 * s_b).
+
+This replaces the round-1 per-block Cholesky + dense-inverse design
+(and removes its 32-TOA-per-epoch GPU cap: the correction is a single
+weighted sum per block at any epoch size).  Numerical safety: the
+subtracted diagonal term is ``beta * d_i^2 <= (1/n_i) * w_i`` with
+``w_i = (1/n_i)/s_b`` the TOA's weight fraction in its block, so the
+cancellation is bounded by the dominance of a single TOA — benign for
+PTA data; validated against the dense oracle in tests at rtol 1e-9.
+
+This module partitions the TOAs into such blocks, permutes them
+contiguous, precomputes (u = 1/nvec, beta), and applies ``N^{-1}`` to
+vectors/panels.
 
 Engine integration (see ``fastfp_amd/engine.py``): with
 ``V = N^{-1} T`` precomputed once, the frequency-domain GEMM is
@@ -90,80 +105,56 @@ class BlockNoise:
 
     # ------------------------------------------------------------------
     def _factor(self):
-        """Cholesky-factor every block and store the dense inverses,
-        packed (CSR-style: values of block b at packed offset
-        ``poff[b] .. poff[b] + sizes[b]^2``, row-major)."""
-        self.poff = np.concatenate(([0], np.cumsum(self.sizes**2)[:-1]))
-        total = int((self.sizes**2).sum())
-        inv = np.empty(total, dtype=np.float64)
-        logdet = 0.0
-        for b in range(len(self.sizes)):
-            s = int(self.sizes[b])
-            o = int(self.offsets[b])
-            blk = np.diag(self.nvec[o : o + s]) + self.ecorr2[b]
-            L = np.linalg.cholesky(blk)
-            logdet += 2.0 * np.log(np.diag(L)).sum()
-            inv[self.poff[b] : self.poff[b] + s * s] = np.linalg.inv(blk).ravel()
-        self.inv_packed = inv
-        self.logdet = logdet
+        """Sherman–Morrison precompute: u = 1/nvec, per-block
+        ``s = sum(u)``, ``beta = e2/(1 + e2*s)``, logdet."""
+        self.uvec = 1.0 / self.nvec
+        s = np.add.reduceat(self.uvec, self.offsets)
+        self.beta = self.ecorr2 / (1.0 + self.ecorr2 * s)
+        self.logdet = float(
+            np.log(self.nvec).sum() + np.log1p(self.ecorr2 * s).sum()
+        )
+        # per-TOA block id for the vectorized segment sums in solve()
+        self.blk_id = np.repeat(
+            np.arange(len(self.sizes), dtype=np.int64), self.sizes
+        )
         self._t = {}
 
     def tensors(self, device):
-        """Device tensors of the packed block data (cached)."""
+        """Device tensors of the block data (cached): u = 1/nvec per
+        TOA, per-block beta, offsets, sizes (int64)."""
         key = str(device)
         if key not in self._t:
             self._t[key] = dict(
-                inv_packed=torch.as_tensor(self.inv_packed, device=device),
+                uvec=torch.as_tensor(self.uvec, device=device),
+                beta=torch.as_tensor(self.beta, device=device),
                 sizes=torch.as_tensor(self.sizes, device=device),
                 offsets=torch.as_tensor(self.offsets, device=device),
-                poff=torch.as_tensor(self.poff, device=device),
+                blk_id=torch.as_tensor(self.blk_id, device=device),
             )
         return self._t[key]
 
-    # ------------------------------------------------------------------
-    def _size_groups(self, device, dtype):
-        """Cached per-(device) grouped gather indices and stacked dense
-        inverse blocks for the batched solve."""
-        key = (str(device), str(dtype))
-        cache = getattr(self, "_groups", None)
-        if cache is None:
-            cache = self._groups = {}
-        if key not in cache:
-            groups = []
-            for s in np.unique(self.sizes):
-                sel = np.nonzero(self.sizes == s)[0]
-                offs = self.offsets[sel]
-                idx = torch.as_tensor(
-                    (offs[:, None] + np.arange(s)[None, :]).ravel(),
-                    device=device,
-                )
-                invs = torch.as_tensor(
-                    np.stack(
-                        [
-                            self.inv_packed[self.poff[b] : self.poff[b] + s * s]
-                            .reshape(s, s)
-                            for b in sel
-                        ]
-                    ),
-                    device=device,
-                    dtype=dtype,
-                )
-                groups.append((int(s), len(sel), idx, invs))
-            cache[key] = groups
-        return cache[key]
-
     def solve(self, X):
         """``N^{-1} X`` for X of shape (ntoa,) or (ntoa, k), in the
-        PERMUTED TOA order.  Torch or numpy in, same type out."""
+        PERMUTED TOA order.  Torch or numpy in, same type out.
+
+        Sherman–Morrison: ``Z = D^{-1}X``; per block,
+        ``out = Z - u * beta_b * (1^T Z)_b`` (one segment sum per
+        column, no dense blocks)."""
         is_np = not isinstance(X, torch.Tensor)
         Xt = torch.as_tensor(np.asarray(X) if is_np else X)
         vec = Xt.dim() == 1
         if vec:
             Xt = Xt[:, None]
-        out = torch.empty_like(Xt)
-        for s, nblk, idx, invs in self._size_groups(Xt.device, Xt.dtype):
-            xb = Xt[idx].reshape(nblk, s, -1)
-            out[idx] = torch.bmm(invs, xb).reshape(nblk * s, -1)
+        bt = self.tensors(Xt.device)
+        u = bt["uvec"].to(Xt.dtype)
+        beta = bt["beta"].to(Xt.dtype)
+        blk_id = bt["blk_id"]
+        Z = Xt * u[:, None]
+        g = torch.zeros(
+            (len(self.sizes), Xt.shape[1]), dtype=Xt.dtype, device=Xt.device
+        )
+        g.index_add_(0, blk_id, Z)
+        out = Z - u[:, None] * (beta[:, None] * g)[blk_id]
         out = out[:, 0] if vec else out
         return out.numpy() if is_np else out
 

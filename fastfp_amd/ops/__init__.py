@@ -121,8 +121,9 @@ def freq_precompute_block(toas, block_noise, Nr, V, TNr, freqs):
 
     ``B = T^T N^-1 S = V^T S`` reuses the sbgemm kernel with V in place
     of T and unit weights; the per-frequency quadratics go through the
-    sigdots_block kernel with the dense per-epoch inverse blocks
-    (factored on device by blockchol_inv).
+    sigdots_block kernel, which applies the exact Sherman–Morrison
+    rank-1 correction per epoch block (any epoch size — no cap; the
+    per-epoch blocks are diagonal + rank-1, see fastfp_amd.blocknoise).
     """
     ext = _try_load()
     F = int(freqs.shape[0])
@@ -131,25 +132,10 @@ def freq_precompute_block(toas, block_noise, Nr, V, TNr, freqs):
     device = V.device
     freqs = freqs.contiguous()
 
-    if block_noise.max_block > 32:
-        raise NotImplementedError(
-            f"GPU block-noise path supports epochs of <= 32 TOAs "
-            f"(got {block_noise.max_block}); use the CPU engine"
-        )
     bt = block_noise.tensors(device)
-    # factor + invert the blocks ON DEVICE (block-Cholesky kernel);
-    # validated against the host factorization in tests
-    nvec = torch.as_tensor(block_noise.nvec, dtype=torch.float64, device=device)
-    ecorr2 = torch.as_tensor(
-        block_noise.ecorr2, dtype=torch.float64, device=device
-    )
-    inv_packed, _logdet = ext.blockchol_inv(
-        nvec, ecorr2, bt["offsets"], bt["sizes"], bt["poff"],
-        int(block_noise.inv_packed.shape[0]),
-    )
     sNs, sNr = ext.sigdots_block(
-        toas.contiguous(), Nr.contiguous(), freqs, inv_packed.contiguous(),
-        bt["offsets"], bt["sizes"], bt["poff"],
+        toas.contiguous(), bt["uvec"].contiguous(), Nr.contiguous(), freqs,
+        bt["beta"].contiguous(), bt["offsets"], bt["sizes"],
     )
 
     RHS = torch.zeros((mp, 2 * F + 1), dtype=torch.float64, device=device)

@@ -18,11 +18,8 @@ void launch_chol_batch(const double*, const double*, int, int, int, int,
 void launch_trsm_fp(const double*, const double*, const double*,
                     const double*, const double*, int, int, int, int,
                     double, double*, hipStream_t);
-void launch_blockchol_inv(const double*, const double*, const long*,
-                          const long*, const long*, int, double*, double*,
-                          hipStream_t);
 void launch_sigdots_block(const double*, const double*, const double*,
-                          const double*, const long*, const long*,
+                          const double*, const double*, const long*,
                           const long*, int, int, int, double*, double*,
                           hipStream_t);
 }
@@ -131,57 +128,36 @@ void trsm_fp_accum(torch::Tensor L, torch::Tensor invd, torch::Tensor RHS,
                  fp.data_ptr<double>(), stream());
 }
 
-// blockchol_inv: factor+invert the per-epoch noise blocks on device.
-// nvec (ntoa, permuted), ecorr2 (nblk), offsets/sizes/poff (int64).
-std::vector<torch::Tensor> blockchol_inv(torch::Tensor nvec,
-                                         torch::Tensor ecorr2,
+// sigdots_block: Sherman-Morrison block-diagonal-N dots.
+// toas/uvec/nr (ntoa, permuted), beta (nblk), offsets/sizes (int64).
+std::vector<torch::Tensor> sigdots_block(torch::Tensor toas, torch::Tensor uvec,
+                                         torch::Tensor nr, torch::Tensor freqs,
+                                         torch::Tensor beta,
                                          torch::Tensor offsets,
-                                         torch::Tensor sizes,
-                                         torch::Tensor poff,
-                                         int64_t total_packed) {
-  check_f64(nvec, "nvec");
-  check_f64(ecorr2, "ecorr2");
-  const int nblk = ecorr2.size(0);
-  auto opts = nvec.options();
-  auto inv = torch::empty({total_packed}, opts);
-  auto logdet = torch::empty({nblk}, opts);
-  launch_blockchol_inv(nvec.data_ptr<double>(), ecorr2.data_ptr<double>(),
-                       offsets.data_ptr<long>(), sizes.data_ptr<long>(),
-                       poff.data_ptr<long>(), nblk, inv.data_ptr<double>(),
-                       logdet.data_ptr<double>(), stream());
-  return {inv, logdet};
-}
-
-std::vector<torch::Tensor> sigdots_block(torch::Tensor toas, torch::Tensor nr,
-                                         torch::Tensor freqs,
-                                         torch::Tensor inv_packed,
-                                         torch::Tensor offsets,
-                                         torch::Tensor sizes,
-                                         torch::Tensor poff) {
+                                         torch::Tensor sizes) {
   check_f64(toas, "toas");
+  check_f64(uvec, "uvec");
   check_f64(nr, "nr");
   check_f64(freqs, "freqs");
-  check_f64(inv_packed, "inv_packed");
+  check_f64(beta, "beta");
   const int ntoa = toas.size(0);
   const int F = freqs.size(0);
   const int nblk = sizes.size(0);
   auto opts = toas.options();
   auto sNs = torch::empty({3, F}, opts);
   auto sNr = torch::empty({2, F}, opts);
-  launch_sigdots_block(toas.data_ptr<double>(), nr.data_ptr<double>(),
-                       freqs.data_ptr<double>(), inv_packed.data_ptr<double>(),
-                       offsets.data_ptr<long>(), sizes.data_ptr<long>(),
-                       poff.data_ptr<long>(), nblk, ntoa, F,
+  launch_sigdots_block(toas.data_ptr<double>(), uvec.data_ptr<double>(),
+                       nr.data_ptr<double>(), freqs.data_ptr<double>(),
+                       beta.data_ptr<double>(), offsets.data_ptr<long>(),
+                       sizes.data_ptr<long>(), nblk, ntoa, F,
                        sNs.data_ptr<double>(), sNr.data_ptr<double>(),
                        stream());
   return {sNs, sNr};
 }
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
-  m.def("blockchol_inv", &blockchol_inv,
-        "per-epoch block Cholesky factor + dense inverse");
   m.def("sigdots_block", &sigdots_block,
-        "fused sincos dots with block-diagonal N");
+        "fused sincos dots with block-diagonal N (Sherman-Morrison)");
   m.def("sigdots", &sigdots, "fused sincos signal-basis dots");
   m.def("sbgemm", &sbgemm, "fused signal-basis MFMA fp64 DGEMM");
   m.def("chol_batch", &chol_batch, "batched LDS-resident fp64 Cholesky");

@@ -612,127 +612,59 @@ __global__ __launch_bounds__(512, 4) void trsm_fp_kernel(
 // implemented this case (/root/reference/fastfp/utils.py:30-31,
 // README.md:22).  Blocks are contiguous after the BlockNoise TOA
 // permutation (fastfp_amd/blocknoise.py).
+//
+// Each block is diag(nvec) + e2*J (diagonal + rank-1), so the inverse
+// is the Sherman-Morrison closed form
+//   N_b^{-1} = D^{-1} - beta_b d d^T,  d = D^{-1} 1,
+//   beta_b = e2 / (1 + e2 * sum(1/nvec_b))
+// — no factorization, O(sz) per block, ANY epoch size (the round-1
+// per-block Cholesky kernel and its 32-TOA cap are gone).
 // ---------------------------------------------------------------------
-#define BLK_MAX 32  // max epoch size handled on-GPU (CPU path beyond)
-
-// blockchol_inv: per block, build N_b = diag(nvec_b) + ecorr2_b * J,
-// Cholesky-factor it and write the dense inverse N_b^{-1} = X^T X
-// (X = L^{-1}), packed row-major at poff[b].  One WAVE per block,
-// register-resident rows, cross-lane shfl (width 32) — the same
-// technique as chol_batch's diagonal factor.
-// grid.x = ceil(nblk / 8), block = 512 (8 waves)
-extern "C" __global__ __launch_bounds__(512) void blockchol_inv_kernel(
-    const double* __restrict__ nvec /*(ntoa,) permuted*/,
-    const double* __restrict__ ecorr2 /*(nblk,)*/,
-    const long* __restrict__ offsets, const long* __restrict__ sizes,
-    const long* __restrict__ poff, int nblk,
-    double* __restrict__ inv_packed, double* __restrict__ logdet_blk) {
-  const int b = blockIdx.x * 8 + (threadIdx.x >> 6);
-  if (b >= nblk) return;
-  const int lane = threadIdx.x & 63;
-  const int i = lane & 31;  // row owned by this lane (2x redundant)
-  const int sz = (int)sizes[b];
-  const long o = offsets[b];
-  const double e2 = ecorr2[b];
-
-  double row[BLK_MAX];  // row i of the block, then of L
-#pragma unroll
-  for (int c = 0; c < BLK_MAX; ++c) {
-    double v = (c < sz && i < sz) ? e2 : 0.0;
-    if (c == i && i < sz) v += nvec[o + i];
-    row[c] = v;
-  }
-  // Cholesky, lanes as rows (width-32 shfl)
-  double ld = 0.0;
-#pragma unroll
-  for (int t = 0; t < BLK_MAX; ++t) {
-    if (t >= sz) break;
-    const double att = __shfl(row[t], t, 32);
-    const double dv = sqrt(att);
-    const double rdv = 1.0 / dv;
-    if (i == t) { row[t] = dv; ld += 2.0 * log(dv); }
-    else if (i > t) row[t] *= rdv;
-#pragma unroll
-    for (int j = t + 1; j < BLK_MAX; ++j) {
-      const double ljt = __shfl(row[t], j, 32);
-      if (j < sz && i >= j) row[j] = fma(-row[t], ljt, row[j]);
-    }
-  }
-  // X = L^{-1}: lane holds column c = i
-  const int c = i;
-  double diag[BLK_MAX], x[BLK_MAX];
-  // one reciprocal per lane, broadcast by shfl (predicated select
-  // avoids a dynamic register index)
-  double dii = 1.0;
-#pragma unroll
-  for (int r = 0; r < BLK_MAX; ++r)
-    if (r == i && r < sz) dii = row[r];
-  const double myrcp = (i < sz) ? 1.0 / dii : 0.0;
-#pragma unroll
-  for (int r = 0; r < BLK_MAX; ++r)
-    diag[r] = (r < sz) ? __shfl(myrcp, r, 32) : 0.0;
-#pragma unroll
-  for (int r = 0; r < BLK_MAX; ++r) {
-    double acc = 0.0;
-#pragma unroll
-    for (int t = 0; t < BLK_MAX; ++t) {
-      const double lrt = __shfl(row[t], r, 32);
-      if (t >= c && t < r && t < sz) acc = fma(lrt, x[t], acc);
-    }
-    x[r] = (r < c) ? 0.0 : (r == c) ? diag[r] : -acc * diag[r];
-  }
-  // N^{-1}[i2][c] = sum_t X[t][i2] * X[t][c]; lane c writes column c
-  double* out = inv_packed + poff[b];
-#pragma unroll
-  for (int i2 = 0; i2 < BLK_MAX; ++i2) {
-    if (i2 >= sz) break;
-    double acc = 0.0;
-#pragma unroll
-    for (int t = 0; t < BLK_MAX; ++t) {
-      const double xti = __shfl(x[t], i2, 32);
-      if (t < sz) acc = fma(xti, x[t], acc);
-    }
-    if (c < sz && lane < 32) out[(long)i2 * sz + c] = acc;
-  }
-  // logdet: each lane contributed its own diagonal's 2*log(dv)
-  for (int off = 16; off > 0; off >>= 1) ld += __shfl_down(ld, off, 32);
-  if (lane == 0) logdet_blk[b] = ld;
-}
 
 // sigdots_block: the five per-frequency dots with BLOCK-diagonal N.
-// sr/cr use the precomputed nr = N^{-1} r vector; the quadratics use
-// the packed dense block inverses.  grid.x = F, block = 256.
+// sr/cr use the precomputed nr = N^{-1} r vector (host S-M solve); the
+// quadratics are the diagonal part (uvec-weighted, like sigdots) minus
+// the per-block rank-1 corrections beta_b * (d.s)_b (d.c)_b.
+// grid.x = F, block = 256.
 extern "C" __global__ __launch_bounds__(256) void sigdots_block_kernel(
-    const double* __restrict__ toas, const double* __restrict__ nr,
-    const double* __restrict__ freqs,
-    const double* __restrict__ inv_packed,
-    const long* __restrict__ offsets, const long* __restrict__ sizes,
-    const long* __restrict__ poff, int nblk, int ntoa, int F,
+    const double* __restrict__ toas, const double* __restrict__ uvec,
+    const double* __restrict__ nr, const double* __restrict__ freqs,
+    const double* __restrict__ beta, const long* __restrict__ offsets,
+    const long* __restrict__ sizes, int nblk, int ntoa, int F,
     double* __restrict__ sNs, double* __restrict__ sNr) {
   const int f = blockIdx.x;
   if (f >= F) return;
   const double w = 2.0 * M_PI * freqs[f];
   double ss = 0, cc = 0, sc = 0, sr = 0, cr = 0;
+  // diagonal part + the N^-1 r dots
   for (int i = threadIdx.x; i < ntoa; i += blockDim.x) {
     double s, c;
     sincos(w * toas[i], &s, &c);
-    sr = fma(s, nr[i], sr);
-    cr = fma(c, nr[i], cr);
+    const double ui = uvec[i];
+    const double ri = nr[i];
+    ss = fma(s * s, ui, ss);
+    cc = fma(c * c, ui, cc);
+    sc = fma(s * c, ui, sc);
+    sr = fma(s, ri, sr);
+    cr = fma(c, ri, cr);
   }
+  // rank-1 corrections: one weighted sum per block (any size)
   for (int b = threadIdx.x; b < nblk; b += blockDim.x) {
-    const int sz = (int)sizes[b];
+    const double bb = beta[b];
+    if (bb == 0.0) continue;  // singleton / no-ecorr block
     const long o = offsets[b];
-    const double* Binv = inv_packed + poff[b];
-    double sv[BLK_MAX], cv[BLK_MAX];
-    for (int i = 0; i < sz; ++i) sincos(w * toas[o + i], &sv[i], &cv[i]);
+    const int sz = (int)sizes[b];
+    double us = 0.0, uc = 0.0;
     for (int i = 0; i < sz; ++i) {
-      for (int j = 0; j < sz; ++j) {
-        const double wij = Binv[(long)i * sz + j];
-        ss = fma(sv[i] * wij, sv[j], ss);
-        cc = fma(cv[i] * wij, cv[j], cc);
-        sc = fma(sv[i] * wij, cv[j], sc);
-      }
+      double s, c;
+      sincos(w * toas[o + i], &s, &c);
+      const double ui = uvec[o + i];
+      us = fma(s, ui, us);
+      uc = fma(c, ui, uc);
     }
+    ss = fma(-bb * us, us, ss);
+    cc = fma(-bb * uc, uc, cc);
+    sc = fma(-bb * us, uc, sc);
   }
   __shared__ double red[4][5];
   ss = wave_reduce_sum(ss);
@@ -830,22 +762,13 @@ void launch_trsm_fp(const double* L, const double* invd, const double* RHS,
   }
 }
 
-void launch_blockchol_inv(const double* nvec, const double* ecorr2,
-                          const long* offsets, const long* sizes,
-                          const long* poff, int nblk, double* inv_packed,
-                          double* logdet_blk, hipStream_t stream) {
-  hipLaunchKernelGGL(blockchol_inv_kernel, dim3((nblk + 7) / 8), dim3(512),
-                     0, stream, nvec, ecorr2, offsets, sizes, poff, nblk,
-                     inv_packed, logdet_blk);
-}
-
-void launch_sigdots_block(const double* toas, const double* nr,
-                          const double* freqs, const double* inv_packed,
-                          const long* offsets, const long* sizes,
-                          const long* poff, int nblk, int ntoa, int F,
+void launch_sigdots_block(const double* toas, const double* uvec,
+                          const double* nr, const double* freqs,
+                          const double* beta, const long* offsets,
+                          const long* sizes, int nblk, int ntoa, int F,
                           double* sNs, double* sNr, hipStream_t stream) {
   hipLaunchKernelGGL(sigdots_block_kernel, dim3(F), dim3(256), 0, stream,
-                     toas, nr, freqs, inv_packed, offsets, sizes, poff,
+                     toas, uvec, nr, freqs, beta, offsets, sizes,
                      nblk, ntoa, F, sNs, sNr);
 }
 

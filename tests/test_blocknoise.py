@@ -48,6 +48,44 @@ def test_blocknoise_solve_vs_dense():
     assert bn.quad(x, y) == pytest.approx(float(x @ np.linalg.solve(N, y)), rel=1e-9)
 
 
+def test_blocknoise_solve_vs_dense_large_epochs():
+    """Sherman–Morrison solve on epochs far beyond 32 TOAs (the old
+    GPU block-Cholesky cap) and with heterogeneous toaerrs — exactness
+    against the dense oracle."""
+    rng = np.random.default_rng(7)
+    psrs = make_synthetic_pta(npsr=1, ntoa=350, tspan_yr=0.02, ntm=3, seed=7)
+    psr = psrs[0]
+    psr.toaerrs = rng.uniform(0.3e-6, 3e-6, psr.ntoa)
+    noise = {}
+    for b in np.unique(psr.backend_flags):
+        noise[f"{psr.name}_basis_ecorr_{b}_log10_ecorr"] = -6.0
+    bn = BlockNoise(psr, noise)
+    assert bn.max_block > 32
+    N = bn.dense()
+    X = rng.normal(size=(psr.ntoa, 4))
+    got = bn.solve(X)
+    want = np.linalg.solve(N, X)
+    np.testing.assert_allclose(got, want, rtol=1e-9, atol=1e-4)
+    assert bn.logdet == pytest.approx(float(np.linalg.slogdet(N)[1]), rel=1e-12)
+
+
+def test_blocknoise_solve_extreme_ecorr():
+    """Stress the S-M cancellation bound: ECORR variance 1e6x the
+    white-noise level (beta*d_i^2 approaches 1/n_i)."""
+    rng = np.random.default_rng(8)
+    psrs = make_synthetic_pta(npsr=1, ntoa=120, tspan_yr=0.01, ntm=3, seed=8)
+    psr = psrs[0]
+    noise = {}
+    for b in np.unique(psr.backend_flags):
+        noise[f"{psr.name}_basis_ecorr_{b}_log10_ecorr"] = -3.0  # huge
+    bn = BlockNoise(psr, noise)
+    N = bn.dense()
+    X = rng.normal(size=(psr.ntoa, 2))
+    got = bn.solve(X)
+    want = np.linalg.solve(N, X)
+    np.testing.assert_allclose(got, want, rtol=1e-7, atol=1e-2)
+
+
 def test_get_xcy_blocknoise_vs_dense_oracle():
     psrs, noise = _psr_and_noise(seed=2)
     psr = psrs[0]

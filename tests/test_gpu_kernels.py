@@ -204,30 +204,34 @@ def test_native_extension_is_loaded():
     assert "fastfp_amd" in _fastfp_hip.__file__
 
 
-def test_blockchol_inv_kernel_vs_host():
-    """On-device per-epoch block Cholesky+inverse vs the host (numpy)
-    factorization."""
+def test_sigdots_block_kernel_vs_eager():
+    """Sherman–Morrison block-N dots kernel vs the CPU eager precompute
+    — on a PTA with LARGE epochs (>32 TOAs/epoch, beyond the round-1
+    block-Cholesky cap)."""
     from fastfp_amd import make_synthetic_pta
     from fastfp_amd.blocknoise import BlockNoise
+    from fastfp_amd.engine import FpEngine
 
-    psrs = make_synthetic_pta(npsr=1, ntoa=300, ntm=3, seed=11)
+    # dense-in-time TOAs -> day-buckets of ~40 TOAs
+    psrs = make_synthetic_pta(npsr=1, ntoa=300, tspan_yr=0.02, ntm=3, seed=11)
     psr = psrs[0]
     noise = {}
     for b in np.unique(psr.backend_flags):
         noise[f"{psr.name}_basis_ecorr_{b}_log10_ecorr"] = -6.4
     bn = BlockNoise(psr, noise)
-    ext = _ext()
-    bt = bn.tensors(DEV)
-    nvec = torch.as_tensor(bn.nvec, dtype=torch.float64, device=DEV)
-    ecorr2 = torch.as_tensor(bn.ecorr2, dtype=torch.float64, device=DEV)
-    inv, logdet = ext.blockchol_inv(
-        nvec, ecorr2, bt["offsets"], bt["sizes"], bt["poff"],
-        int(bn.inv_packed.shape[0]),
-    )
-    np.testing.assert_allclose(
-        inv.cpu().numpy(), bn.inv_packed, rtol=1e-9, atol=1e-30
-    )
-    assert logdet.sum().item() == pytest.approx(bn.logdet, rel=1e-10)
+    assert bn.max_block > 32, "test needs an epoch beyond the old cap"
+    freqs = np.linspace(4e-9, 6e-8, 11)
+    T = np.asarray(psr.Mmat, dtype=np.float64)
+    eng_gpu = FpEngine([psr], [bn], [T], device=DEV)
+    eng_gpu.precompute(freqs)
+    eng_cpu = FpEngine([psr], [bn], [T], device="cpu")
+    eng_cpu.precompute(freqs)
+    for attr in ("sNs", "sNr", "RHS"):
+        got = getattr(eng_gpu.blocks[0], attr).cpu().numpy()
+        want = getattr(eng_cpu.blocks[0], attr).numpy()
+        got = got[: want.shape[0]]  # GPU RHS rows padded to mp
+        np.testing.assert_allclose(got, want, rtol=1e-9, atol=1e-24,
+                                   err_msg=attr)
 
 
 def test_engine_blocknoise_gpu_matches_cpu():
