@@ -607,6 +607,166 @@ __global__ __launch_bounds__(512, 4) void trsm_fp_kernel(
 }
 
 // ---------------------------------------------------------------------
+// trsm_fp_rl: RIGHT-LOOKING variant of trsm_fp.  The left-looking form
+// accumulates each row block's correction in one (two, even/odd-split)
+// dependent MFMA chain; at 64-cycle f64 MFMA dependent latency that
+// caps the pipe near 55% (docs/TUNING_NOTES.md).  Right-looking
+// applies each solved block's update to ALL remaining row blocks
+// immediately:
+//   solve  W[cb] = Iv[cb] (x) W[cb]          (4-MFMA chain)
+//   update W[rb] -= L[rb][cb] (x) W[cb]      (independent chains, one
+//                                             per remaining row block)
+// Same MFMA count, same register state (W only — the separate acc
+// chains vanish, saving 8 VGPRs), but most MFMAs now sit on
+// independent accumulators, so the pipe can fill while the critical
+// solve chain waits.  L is staged as COLUMN panels (double-buffered,
+// row stride 19: conflict-free for the a-fragment access pattern
+// 19*li + 4*kk + lk mod 32 banks).  Selected by FASTFP_TRSM_ALGO=rl.
+// ---------------------------------------------------------------------
+template <int NBT>
+__global__ __launch_bounds__(512, 4) void trsm_fp_rl_kernel(
+    const double* __restrict__ L_all /*(P*D,mp,mp)*/,
+    const double* __restrict__ invd_all /*(P*D, mp/16, 16, 16)*/,
+    const double* __restrict__ RHS_all /*(P, mp, 2F+1)*/,
+    const double* __restrict__ sNs_all /*(P,3,F)*/,
+    const double* __restrict__ sNr_all /*(P,2,F)*/, int F, int D,
+    double gsign, double* __restrict__ fp_all /*(P,D,F)*/) {
+  constexpr int mp = NBT * 16;
+  constexpr int PROWS = NBT > 1 ? (NBT - 1) * 16 : 1;
+  __shared__ double Lc[2][PROWS][19];  // column panel, double-buffered
+  __shared__ double Iv[NBT][16][19];
+  __shared__ double Wu[NBT * 16];  // the solved u column
+
+  const int pp = blockIdx.z;
+  const double* L = L_all + (long)pp * D * mp * mp;
+  const double* invd = invd_all + (long)pp * D * NBT * 256;
+  const double* RHS = RHS_all + (long)pp * mp * (2L * F + 1);
+  const double* sNs = sNs_all + (long)pp * 3 * F;
+  const double* sNr = sNr_all + (long)pp * 2 * F;
+  double* fp = fp_all + (long)pp * D * F;
+  const int d0 = blockIdx.y;
+  const int f0 = blockIdx.x * FPT_FREQS;
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wv = tid >> 6;
+  const long ldr = 2L * F + 1;
+  const int jw = wv * 16;
+  const int li = lane & 15;
+  const int lk = lane >> 4;
+
+  // stage inverted diagonal blocks
+  for (int idx = tid; idx < NBT * 256; idx += 512)
+    Iv[idx >> 8][(idx >> 4) & 15][idx & 15] =
+        invd[(long)d0 * NBT * 256 + idx];
+
+  // load RHS strip into registers (acc layout)
+  const int bc = jw + li;
+  f64x4 W[NBT];
+#pragma unroll
+  for (int rt = 0; rt < NBT; ++rt) {
+#pragma unroll
+    for (int v = 0; v < 4; ++v) {
+      const long row = rt * 16 + 4 * v + lk;
+      double val = 0.0;
+      if (bc < 126) {
+        const long gc = 2L * f0 + bc;
+        if (gc < 2L * F) val = RHS[row * ldr + gc];
+      } else if (bc == 126) {
+        val = RHS[row * ldr + 2L * F];  // the u column
+      }
+      W[rt][v] = val;
+    }
+  }
+
+  // stage column panel 0 (rows 16..mp of column block 0)
+  if (NBT > 1) {
+    for (int idx = tid; idx < (NBT - 1) * 16 * 16; idx += 512) {
+      const int r = idx >> 4, c = idx & 15;
+      Lc[0][r][c] = L[((long)d0 * mp + 16 + r) * mp + c];
+    }
+  }
+  __syncthreads();
+
+#pragma unroll
+  for (int cb = 0; cb < NBT; ++cb) {
+    // solve W[cb] = Iv[cb] (x) W[cb]
+    f64x4 sol = f64x4{0, 0, 0, 0};
+#pragma unroll
+    for (int kk = 0; kk < 4; ++kk) {
+      const double a = Iv[cb][li][kk * 4 + lk];
+      sol = MFMA_F64(a, W[cb][kk], sol);
+    }
+    W[cb] = sol;
+    // independent-accumulator updates of every remaining row block
+#pragma unroll
+    for (int rb = cb + 1; rb < NBT; ++rb) {
+#pragma unroll
+      for (int kk = 0; kk < 4; ++kk) {
+        const double a = -Lc[cb & 1][(rb - cb - 1) * 16 + li][kk * 4 + lk];
+        W[rb] = MFMA_F64(a, sol[kk], W[rb]);
+      }
+    }
+    // prefetch the NEXT column panel (issued after the updates so its
+    // address arithmetic doesn't inflate the MFMA region's live
+    // registers; the loads still overlap the updates' latency since
+    // nothing waits on them until the barrier)
+    if (cb + 1 < NBT) {
+      const int nrows2 = (NBT - 2 - cb) * 16;  // rows (cb+2)*16..mp
+      for (int idx = tid; idx < nrows2 * 16; idx += 512) {
+        const int r = idx >> 4, c = idx & 15;
+        Lc[(cb + 1) & 1][r][c] =
+            L[((long)d0 * mp + (cb + 2) * 16 + r) * mp + (cb + 1) * 16 + c];
+      }
+    }
+    __syncthreads();
+  }
+
+  // fused register/shfl reduction — identical to trsm_fp_kernel
+  if (wv == 7) {
+#pragma unroll
+    for (int rt = 0; rt < NBT; ++rt)
+#pragma unroll
+      for (int v = 0; v < 4; ++v)
+        if (li == 14) Wu[rt * 16 + 4 * v + lk] = W[rt][v];
+  }
+  __syncthreads();
+
+  double pss = 0, pcc = 0, psc = 0, psu = 0, pcu = 0;
+#pragma unroll
+  for (int rt = 0; rt < NBT; ++rt) {
+#pragma unroll
+    for (int v = 0; v < 4; ++v) {
+      const double w = W[rt][v];
+      const double wp = __shfl_xor(w, 1, 64);
+      const double wu = Wu[rt * 16 + 4 * v + lk];
+      pss = fma(w, w, pss);
+      pcc = fma(wp, wp, pcc);
+      psc = fma(w, wp, psc);
+      psu = fma(w, wu, psu);
+      pcu = fma(wp, wu, pcu);
+    }
+  }
+  pss += __shfl_down(pss, 32, 64); pss += __shfl_down(pss, 16, 64);
+  pcc += __shfl_down(pcc, 32, 64); pcc += __shfl_down(pcc, 16, 64);
+  psc += __shfl_down(psc, 32, 64); psc += __shfl_down(psc, 16, 64);
+  psu += __shfl_down(psu, 32, 64); psu += __shfl_down(psu, 16, 64);
+  pcu += __shfl_down(pcu, 32, 64); pcu += __shfl_down(pcu, 16, 64);
+  const int q = (jw + li) >> 1;
+  if (lk == 0 && (li & 1) == 0 && q < FPT_FREQS && f0 + q < F) {
+    const int f = f0 + q;
+    const double M11 = sNs[f] - gsign * pss;
+    const double M22 = sNs[F + f] - gsign * pcc;
+    const double M12 = sNs[2 * F + f] - gsign * psc;
+    const double N1 = sNr[f] - gsign * psu;
+    const double N2 = sNr[F + f] - gsign * pcu;
+    const double det = fma(M11, M22, -M12 * M12);
+    const double num =
+        fma(N1 * N1, M22, fma(-2.0 * N1, N2 * M12, N2 * N2 * M11));
+    fp[(long)d0 * F + f] += 0.5 * num / det;
+  }
+}
+
+// ---------------------------------------------------------------------
 // Block-diagonal white noise (EcorrKernelNoise, BASELINE config 4):
 // N is block-diagonal per observing epoch.  The reference never
 // implemented this case (/root/reference/fastfp/utils.py:30-31,
@@ -735,6 +895,22 @@ void launch_trsm_fp(const double* L, const double* invd, const double* RHS,
   const int ftiles = (F + FPT_FREQS - 1) / FPT_FREQS;
   const dim3 blk(512);
   const int nb = mp >> 4;
+  // FASTFP_TRSM_ALGO=rl selects the right-looking independent-chain
+  // variant (A/B arm; see trsm_fp_rl_kernel)
+  static const char* algo_env = getenv("FASTFP_TRSM_ALGO");
+  static const bool use_rl = algo_env && algo_env[0] == 'r';
+  if (use_rl) {
+    const dim3 grid(ftiles, D, P);
+    switch (nb) {
+#define TRSM_RL_CASE(NBT) \
+      case NBT: hipLaunchKernelGGL((trsm_fp_rl_kernel<NBT>), grid, blk, 0, \
+                    stream, L, invd, RHS, sNs, sNr, F, D, gsign, fp); break;
+      TRSM_RL_CASE(1) TRSM_RL_CASE(2) TRSM_RL_CASE(3) TRSM_RL_CASE(4)
+      TRSM_RL_CASE(5) TRSM_RL_CASE(6) TRSM_RL_CASE(7) TRSM_RL_CASE(8)
+#undef TRSM_RL_CASE
+    }
+    return;
+  }
   // DPG selects draws per workgroup for small solves; the measured
   // default is 1 (8 waves/SIMD at 61 VGPR beats DPG=2's 2 chains at
   // 4 waves/SIMD by ~5%); FASTFP_TRSM_DPG=2 re-enables the A/B arm
