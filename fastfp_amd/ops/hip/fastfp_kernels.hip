@@ -767,6 +767,151 @@ __global__ __launch_bounds__(512, 4) void trsm_fp_rl_kernel(
 }
 
 // ---------------------------------------------------------------------
+// trsm_fp_res: fully LDS-RESIDENT variant.  The r02 PMC profile showed
+// the left-looking and right-looking kernels both stall ~40% of wave
+// cycles on the per-iteration L-panel staging barriers (waitANY 25% +
+// issue waits), NOT on the MFMA dependency chain (the RL experiment
+// measured the chain hypothesis dead: +-2%).  At the compressed shape
+// (NBT<=5) the strictly-lower triangle of L is only NBT*(NBT-1)/2
+// 16x16 tiles (12-24 KB), so the whole factor is staged ONCE at kernel
+// start and the solve loop runs with ZERO barriers and zero memory
+// waits: 2 barriers total per kernel (post-stage, pre-reduction)
+// instead of NBT+1.  Row stride 19 keeps every a-fragment read
+// bank-conflict-free (19*li + 4*kk + lk distinct mod 32).
+// Dispatched for NBT <= 5 when FASTFP_TRSM_ALGO=res (A/B arm).
+// ---------------------------------------------------------------------
+template <int NBT>
+__global__ __launch_bounds__(512, 4) void trsm_fp_res_kernel(
+    const double* __restrict__ L_all /*(P*D,mp,mp)*/,
+    const double* __restrict__ invd_all /*(P*D, mp/16, 16, 16)*/,
+    const double* __restrict__ RHS_all /*(P, mp, 2F+1)*/,
+    const double* __restrict__ sNs_all /*(P,3,F)*/,
+    const double* __restrict__ sNr_all /*(P,2,F)*/, int F, int D,
+    double gsign, double* __restrict__ fp_all /*(P,D,F)*/) {
+  constexpr int mp = NBT * 16;
+  constexpr int NTILE = NBT * (NBT - 1) / 2;  // strictly-lower tiles
+  __shared__ double Lt[NTILE > 0 ? NTILE : 1][16][19];
+  __shared__ double Iv[NBT][16][19];
+  __shared__ double Wu[NBT * 16];
+
+  const int pp = blockIdx.z;
+  const double* L = L_all + (long)pp * D * mp * mp;
+  const double* invd = invd_all + (long)pp * D * NBT * 256;
+  const double* RHS = RHS_all + (long)pp * mp * (2L * F + 1);
+  const double* sNs = sNs_all + (long)pp * 3 * F;
+  const double* sNr = sNr_all + (long)pp * 2 * F;
+  double* fp = fp_all + (long)pp * D * F;
+  const int d0 = blockIdx.y;
+  const int f0 = blockIdx.x * FPT_FREQS;
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wv = tid >> 6;
+  const long ldr = 2L * F + 1;
+  const int jw = wv * 16;
+  const int li = lane & 15;
+  const int lk = lane >> 4;
+
+  // stage EVERYTHING once: inverted diagonal blocks + the strictly-
+  // lower L tiles (tile t(rb,cb) = rb*(rb-1)/2 + cb)
+  for (int idx = tid; idx < NBT * 256; idx += 512)
+    Iv[idx >> 8][(idx >> 4) & 15][idx & 15] =
+        invd[(long)d0 * NBT * 256 + idx];
+  if (NTILE > 0) {
+    for (int idx = tid; idx < NTILE * 256; idx += 512) {
+      const int t = idx >> 8, r = (idx >> 4) & 15, c = idx & 15;
+      // invert t -> (rb, cb): rb = largest with rb*(rb-1)/2 <= t
+      int rb = 1;
+      while (rb * (rb + 1) / 2 <= t) ++rb;
+      const int cb = t - rb * (rb - 1) / 2;
+      Lt[t][r][c] = L[((long)d0 * mp + rb * 16 + r) * mp + cb * 16 + c];
+    }
+  }
+
+  // load RHS strip into registers (acc layout)
+  const int bc = jw + li;
+  f64x4 W[NBT];
+#pragma unroll
+  for (int rt = 0; rt < NBT; ++rt) {
+#pragma unroll
+    for (int v = 0; v < 4; ++v) {
+      const long row = rt * 16 + 4 * v + lk;
+      double val = 0.0;
+      if (bc < 126) {
+        const long gc = 2L * f0 + bc;
+        if (gc < 2L * F) val = RHS[row * ldr + gc];
+      } else if (bc == 126) {
+        val = RHS[row * ldr + 2L * F];  // the u column
+      }
+      W[rt][v] = val;
+    }
+  }
+  __syncthreads();  // everything staged; solve loop is barrier-free
+
+#pragma unroll
+  for (int cb = 0; cb < NBT; ++cb) {
+    f64x4 sol = f64x4{0, 0, 0, 0};
+#pragma unroll
+    for (int kk = 0; kk < 4; ++kk) {
+      const double a = Iv[cb][li][kk * 4 + lk];
+      sol = MFMA_F64(a, W[cb][kk], sol);
+    }
+    W[cb] = sol;
+#pragma unroll
+    for (int rb = cb + 1; rb < NBT; ++rb) {
+#pragma unroll
+      for (int kk = 0; kk < 4; ++kk) {
+        const double a = -Lt[rb * (rb - 1) / 2 + cb][li][kk * 4 + lk];
+        W[rb] = MFMA_F64(a, sol[kk], W[rb]);
+      }
+    }
+  }
+
+  // fused register/shfl reduction — identical to trsm_fp_kernel
+  if (wv == 7) {
+#pragma unroll
+    for (int rt = 0; rt < NBT; ++rt)
+#pragma unroll
+      for (int v = 0; v < 4; ++v)
+        if (li == 14) Wu[rt * 16 + 4 * v + lk] = W[rt][v];
+  }
+  __syncthreads();
+
+  double pss = 0, pcc = 0, psc = 0, psu = 0, pcu = 0;
+#pragma unroll
+  for (int rt = 0; rt < NBT; ++rt) {
+#pragma unroll
+    for (int v = 0; v < 4; ++v) {
+      const double w = W[rt][v];
+      const double wp = __shfl_xor(w, 1, 64);
+      const double wu = Wu[rt * 16 + 4 * v + lk];
+      pss = fma(w, w, pss);
+      pcc = fma(wp, wp, pcc);
+      psc = fma(w, wp, psc);
+      psu = fma(w, wu, psu);
+      pcu = fma(wp, wu, pcu);
+    }
+  }
+  pss += __shfl_down(pss, 32, 64); pss += __shfl_down(pss, 16, 64);
+  pcc += __shfl_down(pcc, 32, 64); pcc += __shfl_down(pcc, 16, 64);
+  psc += __shfl_down(psc, 32, 64); psc += __shfl_down(psc, 16, 64);
+  psu += __shfl_down(psu, 32, 64); psu += __shfl_down(psu, 16, 64);
+  pcu += __shfl_down(pcu, 32, 64); pcu += __shfl_down(pcu, 16, 64);
+  const int q = (jw + li) >> 1;
+  if (lk == 0 && (li & 1) == 0 && q < FPT_FREQS && f0 + q < F) {
+    const int f = f0 + q;
+    const double M11 = sNs[f] - gsign * pss;
+    const double M22 = sNs[F + f] - gsign * pcc;
+    const double M12 = sNs[2 * F + f] - gsign * psc;
+    const double N1 = sNr[f] - gsign * psu;
+    const double N2 = sNr[F + f] - gsign * pcu;
+    const double det = fma(M11, M22, -M12 * M12);
+    const double num =
+        fma(N1 * N1, M22, fma(-2.0 * N1, N2 * M12, N2 * N2 * M11));
+    fp[(long)d0 * F + f] += 0.5 * num / det;
+  }
+}
+
+// ---------------------------------------------------------------------
 // Block-diagonal white noise (EcorrKernelNoise, BASELINE config 4):
 // N is block-diagonal per observing epoch.  The reference never
 // implemented this case (/root/reference/fastfp/utils.py:30-31,
@@ -895,11 +1040,25 @@ void launch_trsm_fp(const double* L, const double* invd, const double* RHS,
   const int ftiles = (F + FPT_FREQS - 1) / FPT_FREQS;
   const dim3 blk(512);
   const int nb = mp >> 4;
-  // FASTFP_TRSM_ALGO=rl selects the right-looking independent-chain
-  // variant (A/B arm; see trsm_fp_rl_kernel)
+  // FASTFP_TRSM_ALGO selects the A/B arm: "rl" = right-looking staged
+  // panels; "res" = fully LDS-resident barrier-free (NBT <= 5, falls
+  // back to rl above that)
   static const char* algo_env = getenv("FASTFP_TRSM_ALGO");
-  static const bool use_rl = algo_env && algo_env[0] == 'r';
-  if (use_rl) {
+  static const bool use_res = algo_env && 0 == __builtin_strcmp(algo_env, "res");
+  static const bool use_rl = algo_env && 0 == __builtin_strcmp(algo_env, "rl");
+  if (use_res && nb <= 5) {
+    const dim3 grid(ftiles, D, P);
+    switch (nb) {
+#define TRSM_RES_CASE(NBT) \
+      case NBT: hipLaunchKernelGGL((trsm_fp_res_kernel<NBT>), grid, blk, 0, \
+                    stream, L, invd, RHS, sNs, sNr, F, D, gsign, fp); break;
+      TRSM_RES_CASE(1) TRSM_RES_CASE(2) TRSM_RES_CASE(3)
+      TRSM_RES_CASE(4) TRSM_RES_CASE(5)
+#undef TRSM_RES_CASE
+    }
+    return;
+  }
+  if (use_rl || (use_res && nb > 5)) {
     const dim3 grid(ftiles, D, P);
     switch (nb) {
 #define TRSM_RL_CASE(NBT) \
