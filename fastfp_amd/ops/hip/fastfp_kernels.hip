@@ -780,7 +780,12 @@ __global__ __launch_bounds__(512, 4) void trsm_fp_rl_kernel(
 // bank-conflict-free (19*li + 4*kk + lk distinct mod 32).
 // Dispatched for NBT <= 5 when FASTFP_TRSM_ALGO=res (A/B arm).
 // ---------------------------------------------------------------------
-template <int NBT>
+// MODE: 0 = full kernel; 1 = skip the reduction/epilogue (timing
+// attribution only — writes nothing); 2 = additionally skip the RHS
+// global loads (pure stage+solve).  Modes 1/2 exist for the phase
+// cost attribution in tools/trsm_diag.py and are never used in
+// production (FASTFP_TRSM_MODE).
+template <int NBT, int MODE = 0>
 __global__ __launch_bounds__(512, 4) void trsm_fp_res_kernel(
     const double* __restrict__ L_all /*(P*D,mp,mp)*/,
     const double* __restrict__ invd_all /*(P*D, mp/16, 16, 16)*/,
@@ -834,13 +839,16 @@ __global__ __launch_bounds__(512, 4) void trsm_fp_res_kernel(
   for (int rt = 0; rt < NBT; ++rt) {
 #pragma unroll
     for (int v = 0; v < 4; ++v) {
-      const long row = rt * 16 + 4 * v + lk;
-      double val = 0.0;
-      if (bc < 126) {
-        const long gc = 2L * f0 + bc;
-        if (gc < 2L * F) val = RHS[row * ldr + gc];
-      } else if (bc == 126) {
-        val = RHS[row * ldr + 2L * F];  // the u column
+      double val = 1.0;
+      if (MODE < 2) {
+        const long row = rt * 16 + 4 * v + lk;
+        val = 0.0;
+        if (bc < 126) {
+          const long gc = 2L * f0 + bc;
+          if (gc < 2L * F) val = RHS[row * ldr + gc];
+        } else if (bc == 126) {
+          val = RHS[row * ldr + 2L * F];  // the u column
+        }
       }
       W[rt][v] = val;
     }
@@ -864,6 +872,15 @@ __global__ __launch_bounds__(512, 4) void trsm_fp_res_kernel(
         W[rb] = MFMA_F64(a, sol[kk], W[rb]);
       }
     }
+  }
+
+  if (MODE >= 1) {
+    // attribution mode: keep the solve live, skip reduction/epilogue
+    double s = 0.0;
+#pragma unroll
+    for (int rt = 0; rt < NBT; ++rt) s += W[rt][0];
+    if (s == 1.2345678e300) fp[0] = s;  // never true for real data
+    return;
   }
 
   // fused register/shfl reduction — identical to trsm_fp_kernel
@@ -1048,6 +1065,19 @@ void launch_trsm_fp(const double* L, const double* invd, const double* RHS,
   static const bool use_rl = algo_env && 0 == __builtin_strcmp(algo_env, "rl");
   if (use_res && nb <= 5) {
     const dim3 grid(ftiles, D, P);
+    // FASTFP_TRSM_MODE=1/2: phase-attribution debug arms (NBT=4 only)
+    static const char* mode_env = getenv("FASTFP_TRSM_MODE");
+    static const int mode = mode_env ? atoi(mode_env) : 0;
+    if (mode == 1 && nb == 4) {
+      hipLaunchKernelGGL((trsm_fp_res_kernel<4, 1>), grid, blk, 0, stream,
+                         L, invd, RHS, sNs, sNr, F, D, gsign, fp);
+      return;
+    }
+    if (mode == 2 && nb == 4) {
+      hipLaunchKernelGGL((trsm_fp_res_kernel<4, 2>), grid, blk, 0, stream,
+                         L, invd, RHS, sNs, sNr, F, D, gsign, fp);
+      return;
+    }
     switch (nb) {
 #define TRSM_RES_CASE(NBT) \
       case NBT: hipLaunchKernelGGL((trsm_fp_res_kernel<NBT>), grid, blk, 0, \
