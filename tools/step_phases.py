@@ -1,0 +1,124 @@
+"""Per-phase timing of one bench step (eager, no hipGraph) at the
+BASELINE config-3 shape — quantifies the non-kernel glue:
+phi assembly, pinv gather/reciprocal, chol, trsm, per-pulsar fp sum.
+
+Run on a GPU box: python tools/step_phases.py [D]
+"""
+
+import os
+import sys
+import time
+
+import numpy as np
+import torch
+
+sys.path.insert(0, os.path.abspath(os.path.join(os.path.dirname(__file__), "..")))
+
+from fastfp_amd import FpEngine, get_mats_nmfp, initialize_pta, make_synthetic_pta
+from fastfp_amd.noise import batch_phiinv, check_batch_homogeneous
+from fastfp_amd import ops
+
+DEV = "cuda:0"
+
+
+def ev_time(fn, iters=10, warmup=3):
+    for _ in range(warmup):
+        fn()
+    torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for _ in range(iters):
+        fn()
+    torch.cuda.synchronize()
+    return (time.perf_counter() - t0) / iters
+
+
+def main():
+    D = int(sys.argv[1]) if len(sys.argv) > 1 else 1000
+    F = 1000
+    psrs = make_synthetic_pta(npsr=67, ntoa=5000, ntm=60, seed=1234, ragged=True)
+    noise = {"gw_gamma": 13.0 / 3.0, "gw_log10_A": float(np.log10(2e-15))}
+    for p in psrs:
+        noise[f"{p.name}_red_noise_gamma"] = 13.0 / 3.0
+        noise[f"{p.name}_red_noise_log10_A"] = -14.5
+    pta = initialize_pta(psrs, noise, inc_cp=True, rn_comps=30, gwb_comps=30)
+    TNTs, Nvecs, Ts = get_mats_nmfp(pta, noise)
+    eng = FpEngine(psrs, Nvecs, Ts, device=DEV)
+    freqs = np.arange(1, F + 1) / pta.Tspan
+    eng.precompute(freqs)
+    eng.enable_draw_compression(
+        [c.var_slice for c in pta.rn_containers],
+        [c.get_phiinv(noise).to(DEV) for c in pta.rn_containers],
+    )
+    for c in pta.rn_containers:
+        c.to(DEV)
+    rng = np.random.default_rng(7)
+    pool = {
+        n: torch.as_tensor(
+            rng.uniform(2, 6, D) if n.endswith("gamma") else rng.uniform(-16, -14, D),
+            dtype=torch.float64, device=DEV,
+        )
+        for n in pta.params
+    }
+    homog = check_batch_homogeneous(pta.rn_containers)
+    fp = torch.zeros((D, F), dtype=torch.float64, device=DEV)
+
+    # full step
+    def full():
+        piv = batch_phiinv(pta.rn_containers, pool, homogeneous=homog)
+        fp.zero_()
+        eng.sweep(phiinvs=piv, draw_chunk=1024, accumulate_to=fp)
+
+    t_full = ev_time(full)
+    print(f"full step        : {t_full*1e3:8.2f} ms   ({D*F/t_full/1e6:.1f}M evals/s)")
+
+    # phase: phi assembly
+    t_phi = ev_time(lambda: batch_phiinv(pta.rn_containers, pool, homogeneous=homog))
+    print(f"phi assembly     : {t_phi*1e3:8.2f} ms")
+
+    piv = batch_phiinv(pta.rn_containers, pool, homogeneous=homog)
+    st = eng._comp_stack
+    P = len(eng.blocks)
+    mv = st["mv"]
+
+    # phase: pinv gather + reciprocal
+    def gather():
+        pin = []
+        for i in range(P):
+            p = piv[i]
+            p = p[None, :] if p.dim() == 1 else p
+            pin.append(p[:, st["vars"][i]])
+        pinv_var = torch.stack(pin)
+        return (1.0 / (pinv_var - st["delta0"][:, None, :])).contiguous()
+
+    t_gather = ev_time(gather)
+    print(f"pinv gather/recip: {t_gather*1e3:8.2f} ms")
+
+    phi_var = gather()
+    mp = ops.pad16(mv)
+
+    from fastfp_amd.ops import _fastfp_hip as ext
+
+    t_chol = ev_time(lambda: ext.chol_batch(st["G"], phi_var, mp))
+    print(f"chol             : {t_chol*1e3:8.2f} ms")
+    L, invd = ext.chol_batch(st["G"], phi_var, mp)
+    pp = torch.zeros((P, D, F), dtype=torch.float64, device=DEV)
+    t_trsm = ev_time(
+        lambda: ext.trsm_fp_accum(L, invd, st["K"], st["M0"], st["N0"], pp, -1.0)
+    )
+    print(f"trsm             : {t_trsm*1e3:8.2f} ms")
+    t_zero = ev_time(lambda: pp.zero_())
+    print(f"pp.zero_         : {t_zero*1e3:8.2f} ms")
+    t_sum = ev_time(lambda: pp.sum(dim=0))
+    print(f"pp.sum(dim=0)    : {t_sum*1e3:8.2f} ms")
+    # alternative layouts for the pulsar sum
+    pp_last = torch.zeros((D, F, P), dtype=torch.float64, device=DEV)
+    t_sum_last = ev_time(lambda: pp_last.sum(dim=-1))
+    print(f"sum last-dim alt : {t_sum_last*1e3:8.2f} ms")
+    acc = sum(
+        (t_phi, t_gather, t_chol, t_trsm, t_zero, t_sum)
+    )
+    print(f"accounted        : {acc*1e3:8.2f} ms of {t_full*1e3:.2f}")
+
+
+if __name__ == "__main__":
+    main()
