@@ -78,6 +78,12 @@ def main():
     piv = batch_phiinv(pta.rn_containers, pool, homogeneous=homog)
     st = eng._comp_stack
     P = len(eng.blocks)
+    if st is None:
+        dropped = [i for i, b in enumerate(eng.blocks) if b.comp is None]
+        mvs = {b.comp["G"].shape[0] for b in eng.blocks if b.comp is not None}
+        kshapes = {tuple(b.comp["K"].shape) for b in eng.blocks if b.comp is not None}
+        print(f"NO comp stack: dropped={dropped} mvs={mvs} kshapes={kshapes}")
+        return
     mv = st["mv"]
 
     # phase: pinv gather + reciprocal
