@@ -383,6 +383,7 @@ class FpEngine:
             # (docs/DESIGN.md §8) and mis-flags healthy models
             scale = fpB.abs().max().clamp_min(1e-30)
             err = (fpA - fpB).abs().max() / scale
+            blk.probe_err = float(err)  # kept for diagnostics
             # NaN-safe: a non-finite probe (GPU kernels don't raise) or
             # a too-large error both disqualify the compressed path
             if not bool(torch.isfinite(err)) or float(err) > tol:
