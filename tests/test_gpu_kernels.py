@@ -429,23 +429,36 @@ def test_probe_keeps_compression_at_bench_conditioning():
     from fastfp_amd import FpEngine, get_mats_nmfp, initialize_pta, \
         make_synthetic_pta
 
-    psrs = make_synthetic_pta(npsr=6, ntoa=3000, tspan_yr=15.0, ntm=8,
-                              seed=1234, ragged=True)
+    # the EXACT bench pulsars that failed the probe in round 2 with the
+    # old jitter_rel=1e-10 (probe errors 1.0-1.8e-5 > tol, conditioning
+    # cond(Sigma_0)*eps — fixed by jitter_rel=1e-8), plus two that
+    # always passed
+    psrs_all = make_synthetic_pta(npsr=67, ntoa=5000, tspan_yr=15.0,
+                                  ntm=60, seed=1234, ragged=True)
+    keep = [0, 1, 10, 13, 20, 26, 66]
     noise = {"gw_gamma": 13.0 / 3.0, "gw_log10_A": float(np.log10(2e-15))}
-    for p in psrs:
+    for p in psrs_all:
         noise[f"{p.name}_red_noise_gamma"] = 13.0 / 3.0
         noise[f"{p.name}_red_noise_log10_A"] = -14.5
-    pta = initialize_pta(psrs, noise, inc_cp=True, rn_comps=30,
-                         gwb_comps=14)
+    pta = initialize_pta(psrs_all, noise, inc_cp=True, rn_comps=30,
+                         gwb_comps=30)
     TNTs, Nvecs, Ts = get_mats_nmfp(pta, noise)
-    eng = FpEngine(psrs, Nvecs, Ts, device=DEV)
-    eng.precompute(np.arange(1, 201) / pta.Tspan)
+    psrs = [psrs_all[i] for i in keep]
+    conts = [pta.rn_containers[i] for i in keep]
+    eng = FpEngine(psrs, [Nvecs[i] for i in keep], [Ts[i] for i in keep],
+                   device=DEV)
+    eng.precompute(np.arange(1, 1001) / pta.Tspan)
     probe = {k: v for k, v in noise.items() if k in pta.params}
     eng.enable_draw_compression(
-        [c.var_slice for c in pta.rn_containers],
-        [c.get_phiinv(probe).to(DEV) for c in pta.rn_containers],
+        [c.var_slice for c in conts],
+        [c.get_phiinv(probe).to(DEV) for c in conts],
     )
-    kept = sum(blk.comp is not None for blk in eng.blocks)
-    assert kept == len(eng.blocks), \
-        f"probe disabled compression on {len(eng.blocks) - kept} pulsars"
+    errs = [getattr(b, "probe_err", float("nan")) for b in eng.blocks]
+    kept_n = sum(blk.comp is not None for blk in eng.blocks)
+    assert kept_n == len(eng.blocks), \
+        f"probe disabled compression on {len(eng.blocks) - kept_n} pulsars " \
+        f"(errors {['%.1e' % e for e in errs]})"
     assert eng._comp_stack is not None, "stacked compressed path must form"
+    # the jitter_rel=1e-8 conditioning fix should leave a wide margin
+    # below the 1e-5 tol, not a marginal pass
+    assert max(errs) < 2e-6, f"probe errors regressed: {errs}"
