@@ -350,11 +350,12 @@ class FpEngine:
             phi_var = (1.0 / (pinv[:, c["var"]] - c["delta0"][None, :])).contiguous()
             try:
                 if self._use_hip and blk.m > 128:
-                    # no direct GPU kernel above m=128: the probe's
-                    # direct reference runs through the CPU LAPACK eager
-                    # path on this one draw (compression precompute is
-                    # already pinned to CPU, so this adds one more small
-                    # host round trip at setup time only)
+                    # above m=128 the GPU direct factor routes through
+                    # rocSOLVER; for the probe's REFERENCE arm prefer
+                    # the CPU LAPACK eager path on this one draw
+                    # (compression precompute is already pinned to CPU,
+                    # so this adds one more small host round trip at
+                    # setup time only)
                     from fastfp_amd import ops
 
                     ops.chol_trsm_fp_accum(
@@ -410,7 +411,7 @@ class FpEngine:
             return
         ms = {blk.m for blk in self.blocks}
         rs = {tuple(blk.RHS.shape) for blk in self.blocks if blk.RHS is not None}
-        if len(ms) != 1 or len(rs) != 1 or self.blocks[0].m > 128:
+        if len(ms) != 1 or len(rs) != 1 or self.blocks[0].m > 256:
             return
         self._direct_stack = dict(
             TNT=torch.stack([b.TNT for b in self.blocks]).contiguous(),

@@ -75,17 +75,18 @@ class FastFp:
     ) -> np.ndarray:
         """Fp over a frequency grid.  Returns (F,) numpy array.
 
-        On GPU with a large basis (m > 128, e.g. GP-ECORR models) the
-        direct per-draw solve kernel does not apply; when the PTAModel
-        is available its variable-bin info routes the sweep through the
-        Schur compression automatically (docs/DESIGN.md §7)."""
+        The direct GPU solve covers basis sizes m <= 256 (rocSOLVER-
+        factored above 128, fastfp_amd.ops.chol_trsm_fp_accum).  For a
+        larger basis (extreme GP-ECORR models) the sweep routes through
+        the Schur compression when the PTAModel's variable-bin info is
+        available, else falls back to the CPU LAPACK engine."""
         if device is None:
             device = "cuda" if torch.cuda.is_available() else "cpu"
         eng = FpEngine(self.psrs, Nvecs, Ts, device=device)
         eng.precompute(freqs, freq_chunk=freq_chunk)
         max_m = max(T.shape[1] for T in Ts)
         conts = getattr(self.pta, "rn_containers", None)
-        if eng._use_hip and max_m > 128 and conts is not None:
+        if eng._use_hip and max_m > 256 and conts is not None:
             phiinvs = [
                 np.diag(np.asarray(sg)) - np.diag(np.asarray(TNT))
                 for sg, TNT in zip(
@@ -95,7 +96,7 @@ class FastFp:
             eng.enable_draw_compression(
                 [c.var_slice for c in conts], phiinvs
             )
-            # the compressed route is the ONLY m>128 GPU path, so guard
+            # the compressed route is the ONLY m>256 GPU path, so guard
             # it twice: the accuracy probe may have dropped pulsars
             # (blk.comp None -> their direct solve would exceed the
             # kernel cap), and near-degenerate priors shrink the
@@ -107,7 +108,7 @@ class FastFp:
                 import warnings
 
                 warnings.warn(
-                    "m > 128 GPU sweep: compression unsafe "
+                    "m > 256 GPU sweep: compression unsafe "
                     f"(margin={margin:.1e}); falling back to the CPU engine"
                 )
                 eng = FpEngine(self.psrs, Nvecs, Ts, device="cpu")
@@ -115,6 +116,16 @@ class FastFp:
                 fp = eng.sweep(sigmas=sigmas)
             else:
                 fp = eng.sweep(phiinvs=phiinvs)
+        elif eng._use_hip and max_m > 256:
+            import warnings
+
+            warnings.warn(
+                "basis size m > 256 without PTAModel variable-bin info; "
+                "falling back to the CPU engine"
+            )
+            eng = FpEngine(self.psrs, Nvecs, Ts, device="cpu")
+            eng.precompute(freqs, freq_chunk=freq_chunk)
+            fp = eng.sweep(sigmas=sigmas)
         else:
             fp = eng.sweep(sigmas=sigmas)
         return fp.cpu().numpy()

@@ -109,12 +109,12 @@ class NMFp:
         if compress and engine.compression_margin(phiinvs) < 1e3:
             # parameter draws too close to the Sigma_0 jitter floor:
             # fall back to the exact direct path
-            if engine._use_hip and any(b.m > 128 for b in engine.blocks):
+            if engine._use_hip and any(b.m > 256 for b in engine.blocks):
                 raise RuntimeError(
                     "noise draws imply red-noise phi so large that the "
                     "compressed solve is numerically unsafe, and the "
                     "direct GPU solve does not support basis size m > "
-                    "128; run these draws on the CPU engine "
+                    "256; run these draws on the CPU engine "
                     "(device='cpu') or restrict the prior range"
                 )
             engine.disable_draw_compression()

@@ -48,7 +48,8 @@ def require_hip() -> None:
         )
 
 
-MAX_MP = 128  # LDS-resident limit of the chol/trsm SOLVE kernels
+MAX_MP_CHOL = 128  # LDS-resident limit of the chol_batch kernel
+MAX_MP = 256  # solve limit: trsm W registers (NBT*4 f64/lane at NBT=16)
 
 
 def pad16(m: int) -> int:
@@ -56,11 +57,12 @@ def pad16(m: int) -> int:
 
 
 def check_m(m: int) -> int:
-    """Solve-dimension check: the batched Cholesky/TRSM kernels keep the
-    matrix (LDS) / RHS strip (registers) on-chip, capped at 128.  The
-    Schur draw compression reduces the per-draw dimension to the
-    variable bins, so large-m models (GP-ECORR) run through NMFp with
-    compression; only the DIRECT per-draw path is capped."""
+    """Solve-dimension check.  Up to mp=128 the batched Cholesky runs
+    in the LDS-resident chol_batch kernel; 128 < mp <= 256 factors via
+    batched rocSOLVER Cholesky + the diag_inv kernel and solves in the
+    same register-resident trsm (W = NBT*4 doubles per lane).  Beyond
+    256 use the Schur-compressed path (per-draw dimension = variable
+    bins) or the CPU engine."""
     mp = pad16(m)
     if mp > MAX_MP:
         raise NotImplementedError(
@@ -154,14 +156,36 @@ def freq_precompute_block(toas, block_noise, Nr, V, TNr, freqs):
 
 
 def chol_trsm_fp_accum(TNT, phiinv, RHS, sNs, sNr, fp_out, gsign=1.0):
-    """Batched Cholesky of Sigma = TNT + diag(phiinv) (assembled in
-    LDS), then the fused triangular solve of RHS (mp, 2F+1) + 2x2 Fp
-    reduction, accumulating into fp_out (D, F).
+    """Batched Cholesky of Sigma = TNT + diag(phiinv), then the fused
+    triangular solve of RHS (mp, 2F+1) + 2x2 Fp reduction, accumulating
+    into fp_out (D, F).
+
+    mp <= 128: LDS-resident chol_batch kernel.  128 < mp <= 256
+    (GP-ECORR direct path): Sigma no longer fits LDS — factor through
+    batched rocSOLVER Cholesky (substitution-based, accurate; the
+    rocBLAS accuracy caveat of docs/TUNING_NOTES.md concerns its
+    inversion-based trsm, not potrf) + the diag_inv kernel, then the
+    same register-resident trsm.
 
     ``gsign``: +1 direct path (M = sNs - W.W); -1 Schur-compressed draw
     path (M = M0 + W.W) — docs/DESIGN.md §draw compression."""
     ext = _try_load()
     m = TNT.shape[-1]
     mp = check_m(m)
-    L, invd = ext.chol_batch(TNT.contiguous(), phiinv.contiguous(), mp)
+    if mp <= MAX_MP_CHOL:
+        L, invd = ext.chol_batch(TNT.contiguous(), phiinv.contiguous(), mp)
+    else:
+        batched = TNT.dim() == 3
+        TNTb = TNT if batched else TNT.unsqueeze(0)  # (P, m, m)
+        pib = phiinv if batched else phiinv.unsqueeze(0)  # (P, D, m)
+        P, D = pib.shape[0], pib.shape[1]
+        sigma = TNTb[:, None, :, :] + torch.diag_embed(pib)  # (P, D, m, m)
+        Lt = torch.linalg.cholesky(sigma.reshape(P * D, m, m))
+        L = torch.zeros((P * D, mp, mp), dtype=TNT.dtype, device=TNT.device)
+        L[:, :m, :m] = Lt
+        # identity padding: pad rows of RHS are zero, so results are
+        # unchanged (same convention as chol_batch)
+        idx = torch.arange(m, mp, device=TNT.device)
+        L[:, idx, idx] = 1.0
+        invd = ext.diag_inv(L.contiguous())
     ext.trsm_fp_accum(L, invd, RHS, sNs, sNr, fp_out, gsign)

@@ -22,6 +22,7 @@ void launch_sigdots_block(const double*, const double*, const double*,
                           const double*, const double*, const long*,
                           const long*, int, int, int, double*, double*,
                           hipStream_t);
+void launch_diag_inv(const double*, int, long, double*, hipStream_t);
 }
 
 namespace {
@@ -121,11 +122,25 @@ void trsm_fp_accum(torch::Tensor L, torch::Tensor invd, torch::Tensor RHS,
               "RHS shape");
   TORCH_CHECK(fp.dim() == (batched ? 3 : 2) && fp.size(batched ? 2 : 1) == F,
               "fp shape");
-  TORCH_CHECK(mp % 16 == 0 && mp <= 128, "mp must be <=128, multiple of 16");
+  TORCH_CHECK(mp % 16 == 0 && mp <= 256, "mp must be <=256, multiple of 16");
   launch_trsm_fp(L.data_ptr<double>(), invd.data_ptr<double>(),
                  RHS.data_ptr<double>(), sNs.data_ptr<double>(),
                  sNr.data_ptr<double>(), mp, F, D, P, gsign,
                  fp.data_ptr<double>(), stream());
+}
+
+// diag_inv: invert the 16x16 diagonal blocks of batched lower-tri L
+// (B, mp, mp) -> invd (B, mp/16, 16, 16); the m > 128 direct path.
+torch::Tensor diag_inv(torch::Tensor L) {
+  check_f64(L, "L");
+  TORCH_CHECK(L.dim() == 3 && L.size(1) == L.size(2), "L must be (B,mp,mp)");
+  const long B = L.size(0);
+  const int mp = L.size(1);
+  TORCH_CHECK(mp % 16 == 0, "mp multiple of 16");
+  auto invd = torch::empty({B, mp / 16, 16, 16}, L.options());
+  launch_diag_inv(L.data_ptr<double>(), mp, B * (mp / 16),
+                  invd.data_ptr<double>(), stream());
+  return invd;
 }
 
 // sigdots_block: Sherman-Morrison block-diagonal-N dots.
@@ -158,6 +173,8 @@ std::vector<torch::Tensor> sigdots_block(torch::Tensor toas, torch::Tensor uvec,
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("sigdots_block", &sigdots_block,
         "fused sincos dots with block-diagonal N (Sherman-Morrison)");
+  m.def("diag_inv", &diag_inv,
+        "invert 16x16 diagonal blocks of batched lower-triangular L");
   m.def("sigdots", &sigdots, "fused sincos signal-basis dots");
   m.def("sbgemm", &sbgemm, "fused signal-basis MFMA fp64 DGEMM");
   m.def("chol_batch", &chol_batch, "batched LDS-resident fp64 Cholesky");

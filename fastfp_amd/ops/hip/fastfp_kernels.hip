@@ -929,6 +929,53 @@ __global__ __launch_bounds__(512, 4) void trsm_fp_res_kernel(
 }
 
 // ---------------------------------------------------------------------
+// diag_inv: invert the 16x16 diagonal blocks of a batch of lower-
+// triangular L (B, mp, mp) -> invd (B, mp/16, 16, 16).  Used by the
+// m > 128 direct path, where L comes from batched rocSOLVER Cholesky
+// (the LDS-resident chol_batch kernel caps at mp = 128: Sigma no
+// longer fits LDS above that).  One wave per (batch, block); register
+// rows + width-16 shfl — same technique as chol_batch's CHOL_DIAG.
+// grid.x = ceil(B * mp/16 / 8), block = 512 (8 waves)
+// ---------------------------------------------------------------------
+extern "C" __global__ __launch_bounds__(512, 4) void diag_inv_kernel(
+    const double* __restrict__ L, int mp, long nblk_total,
+    double* __restrict__ invd) {
+  const long w = (long)blockIdx.x * 8 + (threadIdx.x >> 6);
+  if (w >= nblk_total) return;
+  const int nbt = mp >> 4;
+  const long b = w / nbt;
+  const int kb = (int)(w % nbt);
+  const int lane = threadIdx.x & 63;
+  const int i = lane & 15;  // row; the 4 sub-groups compute redundantly
+  const double* Lb = L + b * (long)mp * mp + (long)kb * 16 * mp + kb * 16;
+  double row[16];
+#pragma unroll
+  for (int c = 0; c < 16; ++c) row[c] = (c <= i) ? Lb[(long)i * mp + c] : 0.0;
+  const int c = i;  // lane owns output column c
+  double x[16];
+  double dii = 1.0;
+#pragma unroll
+  for (int r = 0; r < 16; ++r)
+    if (r == i) dii = row[r];
+  const double myrcp = 1.0 / dii;
+#pragma unroll
+  for (int r = 0; r < 16; ++r) {
+    const double dr = __shfl(myrcp, r, 16);
+    double acc = 0.0;
+#pragma unroll
+    for (int t = 0; t < 16; ++t) {
+      const double lrt = __shfl(row[t], r, 16);
+      if (t >= c && t < r) acc = fma(lrt, x[t], acc);
+    }
+    x[r] = (r < c) ? 0.0 : (r == c) ? dr : -acc * dr;
+  }
+  if (lane < 16)
+#pragma unroll
+    for (int r = 0; r < 16; ++r)
+      invd[(b * nbt + kb) * 256 + r * 16 + c] = x[r];
+}
+
+// ---------------------------------------------------------------------
 // Block-diagonal white noise (EcorrKernelNoise, BASELINE config 4):
 // N is block-diagonal per observing epoch.  The reference never
 // implemented this case (/root/reference/fastfp/utils.py:30-31,
@@ -1088,7 +1135,7 @@ void launch_trsm_fp(const double* L, const double* invd, const double* RHS,
     }
     return;
   }
-  if (use_rl || (use_res && nb > 5)) {
+  if ((use_rl || (use_res && nb > 5)) && nb <= 8) {
     const dim3 grid(ftiles, D, P);
     switch (nb) {
 #define TRSM_RL_CASE(NBT) \
@@ -1123,8 +1170,25 @@ void launch_trsm_fp(const double* L, const double* invd, const double* RHS,
       TRSM_CASE(1) TRSM_CASE(2) TRSM_CASE(3) TRSM_CASE(4)
       TRSM_CASE(5) TRSM_CASE(6) TRSM_CASE(7) TRSM_CASE(8)
 #undef TRSM_CASE
+      // m > 128 (GP-ECORR direct path) dispatches the RIGHT-LOOKING
+      // variant: its single in-place accumulator set (no acc/acc2
+      // chains) keeps W = NBT*4 f64 regs per lane within budget at
+      // NBT up to 16.  The FACTOR for these sizes comes from rocSOLVER
+      // (chol_batch's LDS-resident Sigma caps at 128) + diag_inv.
+#define TRSM_BIG_CASE(NBT) \
+      case NBT: hipLaunchKernelGGL((trsm_fp_rl_kernel<NBT>), grid, blk, 0, \
+                    stream, L, invd, RHS, sNs, sNr, F, D, gsign, fp); break;
+      TRSM_BIG_CASE(9) TRSM_BIG_CASE(10) TRSM_BIG_CASE(11) TRSM_BIG_CASE(12)
+      TRSM_BIG_CASE(13) TRSM_BIG_CASE(14) TRSM_BIG_CASE(15) TRSM_BIG_CASE(16)
+#undef TRSM_BIG_CASE
     }
   }
+}
+
+void launch_diag_inv(const double* L, int mp, long nblk_total, double* invd,
+                     hipStream_t stream) {
+  hipLaunchKernelGGL(diag_inv_kernel, dim3((nblk_total + 7) / 8), dim3(512),
+                     0, stream, L, mp, nblk_total, invd);
 }
 
 void launch_sigdots_block(const double* toas, const double* uvec,

@@ -362,6 +362,54 @@ def test_engine_direct_stacked_gpu_matches_cpu():
     np.testing.assert_allclose(got, want, rtol=1e-8)
 
 
+def test_direct_sweep_large_m_gpu():
+    """DIRECT (uncompressed) GPU sweep at basis size 128 < m <= 256:
+    the rocSOLVER-factored + diag_inv + register-resident trsm path
+    (round-1 cap lifted).  Validated against the CPU eager engine."""
+    from fastfp_amd import FpEngine, get_mats_nmfp, initialize_pta, \
+        make_synthetic_pta
+
+    psrs = make_synthetic_pta(npsr=2, ntoa=500, ntm=10, seed=51,
+                              ragged=False)
+    noise = {"gw_gamma": 13.0 / 3.0, "gw_log10_A": float(np.log10(2e-15))}
+    for p in psrs:
+        noise[f"{p.name}_red_noise_gamma"] = 4.0
+        noise[f"{p.name}_red_noise_log10_A"] = -14.5
+    # ntm 10 + 2*80 rn + 2*10 gwb = 190 columns
+    pta = initialize_pta(psrs, noise, inc_cp=True, rn_comps=80,
+                         gwb_comps=10)
+    TNTs, Nvecs, Ts = get_mats_nmfp(pta, noise)
+    assert 128 < max(T.shape[1] for T in Ts) <= 256
+    D = 4
+    rng = np.random.default_rng(5)
+    pars = {
+        n: (rng.uniform(2, 6, D) if n.endswith("gamma")
+            else rng.uniform(-16, -14, D))
+        for n in pta.params
+    }
+    phiinvs = [c.get_phiinv(pars).numpy() for c in pta.rn_containers]
+    freqs = np.linspace(4e-9, 6e-8, 13)
+
+    eng_c = FpEngine(psrs, Nvecs, Ts, device="cpu").precompute(freqs)
+    want = eng_c.sweep(phiinvs=phiinvs).numpy()
+    eng_g = FpEngine(psrs, Nvecs, Ts, device=DEV).precompute(freqs)
+    got = eng_g.sweep(phiinvs=phiinvs).cpu().numpy()
+    np.testing.assert_allclose(got, want, rtol=1e-7)
+
+    # plain-Fp entry point (sigmas contract) through the same path
+    from fastfp_amd import FastFp, get_mats_fp
+
+    noise1 = dict(noise)
+    for p in psrs:
+        noise1[f"{p.name}_red_noise_gamma"] = 4.2
+        noise1[f"{p.name}_red_noise_log10_A"] = -14.2
+    Nv2, Ts2, sigmas = get_mats_fp(pta, noise1)
+    fp_obj = FastFp(psrs, pta)
+    cpu = fp_obj.sweep(freqs, Nv2, Ts2, sigmas, device="cpu")
+    gpu = fp_obj.sweep(freqs, Nv2, Ts2, sigmas, device=DEV)
+    np.testing.assert_allclose(gpu, cpu, rtol=1e-7)
+
+
 def test_graph_captured_sweep_bitwise_equals_eager():
     """A hipGraph-captured sweep (the bench's step structure) must
     reproduce the eager sweep BITWISE on replay."""
