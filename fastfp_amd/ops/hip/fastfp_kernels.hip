@@ -213,6 +213,10 @@ extern "C" __global__ __launch_bounds__(512) void sbgemm_kernel(
 // (width 16) -- the LDS read-modify-write chains of the naive version
 // were the kernel's dominant cost (profiles/r01_initial_stats.md).
 // ---------------------------------------------------------------------
+// (A (+i)-rotation column swizzle in place of the +1 row pad would
+// shave Ash from 33.3 to 32.0 KB at mp=64 — one more workgroup per CU
+// — but the extra address arithmetic measured +16 VGPR, dropping a
+// waves/SIMD tier: net zero.  Padded layout kept.)
 #define A_(i, j) Ash[(i) * (NBT * 16 + 1) + (j)]
 
 template <int NBT>
@@ -279,22 +283,23 @@ __global__ __launch_bounds__(512, 4) void chol_batch_kernel(
     _Pragma("unroll") for (int c = 0; c < 16; ++c) if (c <= i)               \
         A_(dk0 + i, dk0 + c) = row[c];                                       \
     const int c = i;                                                         \
-    double diag[16], x[16];                                                  \
+    double x[16];                                                            \
     /* lane i computes only 1/L_ii; the rest arrive by shfl (16x less   */   \
     /* serial f64 division work than every lane inverting all 16).      */   \
     /* predicated select avoids a dynamic register index (scratch!)     */   \
+    /* the reciprocal broadcast is inlined per r (a diag[16] register   */   \
+    /* array cost ~16 VGPR and a waves/SIMD tier)                       */   \
     double dii = 0.0;                                                        \
     _Pragma("unroll") for (int r = 0; r < 16; ++r) if (r == i) dii = row[r]; \
     const double myrcp = 1.0 / dii;                                          \
-    _Pragma("unroll") for (int r = 0; r < 16; ++r) diag[r] =                 \
-        __shfl(myrcp, r, 16);                                                \
     _Pragma("unroll") for (int r = 0; r < 16; ++r) {                         \
+      const double dr = __shfl(myrcp, r, 16);                                \
       double acc2 = 0.0;                                                     \
       _Pragma("unroll") for (int t = 0; t < 16; ++t) {                       \
         const double lrt = __shfl(row[t], r, 16);                            \
         if (t >= c && t < r) acc2 = fma(lrt, x[t], acc2);                    \
       }                                                                      \
-      x[r] = (r < c) ? 0.0 : (r == c) ? diag[r] : -acc2 * diag[r];           \
+      x[r] = (r < c) ? 0.0 : (r == c) ? dr : -acc2 * dr;                     \
     }                                                                        \
     _Pragma("unroll") for (int r = 0; r < 16; ++r) (DST)[r][c] = x[r];       \
     if (lane < 16)                                                           \
@@ -347,14 +352,19 @@ __global__ __launch_bounds__(512, 4) void chol_batch_kernel(
     }
     __syncthreads();
 
-    // phase A2: wave 0 factors diagonal kb+1 while waves 1..7 do the
-    // remaining trailing tiles (jb >= kb+2)
-    if (wv == 0) {
+    // phase A2: ONE wave factors diagonal kb+1 while the others do the
+    // remaining trailing tiles (jb >= kb+2).  The factor wave ROTATES
+    // with kb: waves of a workgroup land on SIMD (wv % 4), so a fixed
+    // wave 0 puts every concurrent workgroup's serial diagonal on
+    // SIMD 0 while SIMDs 1-3 starve (measured 82% parked) — rotation
+    // spreads the serial phase across all four SIMDs.
+    const int dw = (kb + 1) % nwv;
+    if (wv == dw) {
       CHOL_DIAG(kb + 1, inv16[(kb + 1) & 1]);
     } else {
       const int t = nb - kb - 2;
       const int ntile = t * (t + 1) / 2;
-      for (int q = wv - 1; q < ntile; q += nwv - 1) {
+      for (int q = wv - (wv > dw ? 1 : 0); q < ntile; q += nwv - 1) {
         int ib = kb + 2, rem = q;
         while (rem > ib - kb - 2) { rem -= (ib - kb - 1); ++ib; }
         const int jb = kb + 2 + rem;
