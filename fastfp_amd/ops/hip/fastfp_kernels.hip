@@ -352,19 +352,17 @@ __global__ __launch_bounds__(512, 4) void chol_batch_kernel(
     }
     __syncthreads();
 
-    // phase A2: ONE wave factors diagonal kb+1 while the others do the
-    // remaining trailing tiles (jb >= kb+2).  The factor wave ROTATES
-    // with kb: waves of a workgroup land on SIMD (wv % 4), so a fixed
-    // wave 0 puts every concurrent workgroup's serial diagonal on
-    // SIMD 0 while SIMDs 1-3 starve (measured 82% parked) — rotation
-    // spreads the serial phase across all four SIMDs.
-    const int dw = (kb + 1) % nwv;
-    if (wv == dw) {
+    // phase A2: wave 0 factors diagonal kb+1 while the other waves do
+    // the remaining trailing tiles (jb >= kb+2).  (Rotating the factor
+    // wave across kb to spread the serial phase over SIMDs was
+    // measured NEUTRAL — the kernel's parking is not SIMD-0 contention
+    // — and was reverted.)
+    if (wv == 0) {
       CHOL_DIAG(kb + 1, inv16[(kb + 1) & 1]);
     } else {
       const int t = nb - kb - 2;
       const int ntile = t * (t + 1) / 2;
-      for (int q = wv - (wv > dw ? 1 : 0); q < ntile; q += nwv - 1) {
+      for (int q = wv - 1; q < ntile; q += nwv - 1) {
         int ib = kb + 2, rem = q;
         while (rem > ib - kb - 2) { rem -= (ib - kb - 1); ++ib; }
         const int jb = kb + 2 + rem;
