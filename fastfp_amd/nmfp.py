@@ -29,6 +29,7 @@ class NMFp:
         self.rn_sigs = rn_sigs
         self.toas = [np.asarray(p.toas, dtype=np.float64) for p in psrs]
         self.residuals = [np.asarray(p.residuals, dtype=np.float64) for p in psrs]
+        self._phi_homog = None  # cached check_batch_homogeneous result
 
     def __call__(self, fgw, samples, Nvecs, Ts, TNTs):
         return self.calculate_nmfp(fgw, samples, Nvecs, Ts, TNTs)
@@ -101,9 +102,14 @@ class NMFp:
                     [sig.var_slice for sig in self.rn_sigs],
                     [sig.get_phiinv(probe) for sig in self.rn_sigs],
                 )
-        from fastfp_amd.noise import batch_phiinv
+        from fastfp_amd.noise import batch_phiinv, check_batch_homogeneous
 
-        phiinvs = batch_phiinv(self.rn_sigs, samples)
+        # the homogeneity check device-syncs; cache it across the
+        # per-batch calls of a long CLI sweep
+        if self._phi_homog is None:
+            self._phi_homog = check_batch_homogeneous(self.rn_sigs)
+        phiinvs = batch_phiinv(self.rn_sigs, samples,
+                               homogeneous=self._phi_homog)
         # scalar-parameter dicts produce (m,) vectors; promote to (1, m)
         phiinvs = [p[None, :] if p.dim() == 1 else p for p in phiinvs]
         if compress and engine.compression_margin(phiinvs) < 1e3:
