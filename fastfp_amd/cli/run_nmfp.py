@@ -222,17 +222,28 @@ def main(
 
     writer = ThreadPoolExecutor(max_workers=1) if checkpoint else None
     pending = []
+    timing = os.environ.get("FASTFP_CLI_TIMING") == "1"
+    tacc = {"map": 0.0, "sweep": 0.0, "save": 0.0}
     for lo in batches:
         sel = my_idx[lo : lo + batch_size]
         ck = os.path.join(batch_dir, f"r{rank}_b{lo}.npy")
         if checkpoint and resume and os.path.exists(ck):
             parts.append(np.load(ck))
             continue
+        tm0 = time.perf_counter()
         samples = map_params(pta, rns_full[:, sel])
+        tm1 = time.perf_counter()
         vals = nmfp.sweep(freqs, samples, Nvecs, Ts, engine=eng)
+        tm2 = time.perf_counter()
         if writer is not None:
             pending.append(writer.submit(np.save, ck, vals))
         parts.append(vals)
+        if timing:
+            tacc["map"] += tm1 - tm0
+            tacc["sweep"] += tm2 - tm1
+    if timing:
+        logger.info(f"CLI timing: map {tacc['map']:.3f} s, "
+                    f"sweep {tacc['sweep']:.3f} s")
     if writer is not None:
         for fut in pending:
             fut.result()  # surface write errors before declaring success
