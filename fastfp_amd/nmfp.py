@@ -13,6 +13,7 @@ costs one Cholesky + one fused triangular-solve/reduction instead of
 from __future__ import annotations
 
 import math
+import os
 
 import numpy as np
 import torch
@@ -124,6 +125,19 @@ class NMFp:
                     "(device='cpu') or restrict the prior range"
                 )
             engine.disable_draw_compression()
+        if os.environ.get("FASTFP_CLI_TIMING") == "1":
+            import time
+
+            torch.cuda.synchronize() if engine.device.type == "cuda" else None
+            t0 = time.perf_counter()
+            fp = engine.sweep(phiinvs=phiinvs, draw_chunk=draw_chunk)
+            torch.cuda.synchronize() if engine.device.type == "cuda" else None
+            t1 = time.perf_counter()
+            out = fp.cpu().numpy()
+            t2 = time.perf_counter()
+            print(f"[nmfp.sweep] engine.sweep {t1-t0:.3f} s, "
+                  f"to-host {t2-t1:.3f} s", flush=True)
+            return out
         fp = engine.sweep(phiinvs=phiinvs, draw_chunk=draw_chunk)
         return fp.cpu().numpy()
 
