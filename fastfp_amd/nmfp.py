@@ -135,8 +135,12 @@ class NMFp:
             t1 = time.perf_counter()
             out = fp.cpu().numpy()
             t2 = time.perf_counter()
+            ncomp = sum(b.comp is not None for b in engine.blocks)
             print(f"[nmfp.sweep] engine.sweep {t1-t0:.3f} s, "
-                  f"to-host {t2-t1:.3f} s", flush=True)
+                  f"to-host {t2-t1:.3f} s; stack="
+                  f"{getattr(engine, '_comp_stack', None) is not None} "
+                  f"comp={ncomp}/{len(engine.blocks)} chunk={draw_chunk}",
+                  flush=True)
             return out
         fp = engine.sweep(phiinvs=phiinvs, draw_chunk=draw_chunk)
         return fp.cpu().numpy()
