@@ -250,9 +250,9 @@ class FpEngine:
             # it): the compression error scales ~ cond(Sigma_0)*eps ~
             # eps/jitter_rel, so 1e-8 keeps errors ~1e-9..1e-7 of the
             # spectrum scale (measured: 100x better than the r01
-            # 1e-10, which left 6/67 bench pulsars over the probe tol)
-            # while the bench-shape margin stays ~1e4 >> the 1e3
-            # fallback threshold.
+            # 1e-10, which left 6/67 bench pulsars over the probe tol).
+            # The margin guard (compression_margin) falls back to the
+            # direct path below margin 1.5.
             # rank-deficient TNT (basis larger than the TOA count)
             # leaves Sigma_0 supported only by the jitter along its
             # null space — keep such pulsars on the exact direct path
@@ -449,8 +449,11 @@ class FpEngine:
 
     def compression_margin(self, phiinvs) -> float:
         """min over pulsars/draws/bins of phiinv_d / delta_0.  The
-        compressed path is numerically safe when this is >> 1 (callers
-        fall back to the direct path below ~1e3).
+        compressed path is numerically safe above ~1.5 (the split is
+        algebraically exact; precision of the Delta = phiinv - delta0
+        subtraction degrades as eps/(1 - 1/margin) and Delta <= 0 is
+        the NaN cliff — measured 2e-10 error at margin 1.7).  Callers
+        fall back to the exact direct path below 1.5.
 
         One device sync total: per-pulsar mins are stacked and reduced
         on device (a per-pulsar ``float()`` cost 67 syncs per draw

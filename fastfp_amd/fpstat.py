@@ -103,8 +103,9 @@ class FastFp:
             # phi/jitter margin the Woodbury correction relies on.  In
             # either case fall back to the CPU LAPACK engine — slower
             # but always exact.
+            # threshold 1.5: see the margin-guard note in nmfp.sweep
             margin = eng.compression_margin(phiinvs)
-            if any(blk.comp is None for blk in eng.blocks) or margin < 1e3:
+            if any(blk.comp is None for blk in eng.blocks) or margin < 1.5:
                 import warnings
 
                 warnings.warn(

@@ -113,8 +113,18 @@ class NMFp:
                                homogeneous=self._phi_homog)
         # scalar-parameter dicts produce (m,) vectors; promote to (1, m)
         phiinvs = [p[None, :] if p.dim() == 1 else p for p in phiinvs]
-        if compress and engine.compression_margin(phiinvs) < 1e3:
-            # parameter draws too close to the Sigma_0 jitter floor:
+        # Margin guard threshold: the compressed path is ALGEBRAICALLY
+        # exact for any Delta_d = phiinv_d - delta0 > 0; the only
+        # numerical hazard is the subtraction's precision loss as
+        # margin -> 1 (rel err = eps/(1 - 1/margin)) and the Delta <= 0
+        # cliff (indefinite C_d -> NaN from the Cholesky solver).
+        # Measured: at margin 1.7 the compressed-vs-direct error is
+        # 2e-10 of the spectrum scale (docs/PERFORMANCE.md r02) — the
+        # round-1 threshold of 1e3 was ~3 orders too conservative and
+        # silently pushed whole uniform-prior sweeps onto the 4x-slower
+        # direct path.
+        if compress and engine.compression_margin(phiinvs) < 1.5:
+            # draws at/below the Sigma_0 jitter floor:
             # fall back to the exact direct path
             if engine._use_hip and any(b.m > 256 for b in engine.blocks):
                 raise RuntimeError(

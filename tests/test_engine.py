@@ -232,6 +232,48 @@ def test_fp_detects_injected_cw_signal():
     assert fp.max() > 50.0, fp.max()
 
 
+def test_compression_low_margin_draws_stay_compressed_and_exact():
+    """Draws whose margin is small-but-safe (>1.5; e.g. uniform-prior
+    tails with large A and gamma) must STAY on the compressed path and
+    agree with the exact direct sweep — the r01 threshold of 1e3
+    silently pushed whole uniform-prior sweeps onto the direct path."""
+    psrs = make_synthetic_pta(npsr=2, ntoa=400, ntm=8, seed=16)
+    noise = {"gw_gamma": 13.0 / 3.0, "gw_log10_A": float(np.log10(2e-15))}
+    for p in psrs:
+        noise[f"{p.name}_red_noise_gamma"] = 13.0 / 3.0
+        noise[f"{p.name}_red_noise_log10_A"] = -14.5
+    pta = initialize_pta(psrs, noise, inc_cp=True, rn_comps=15, gwb_comps=15)
+    TNTs, Nvecs, Ts = get_mats_nmfp(pta, noise)
+    D = 4
+    samples = {
+        n: (np.array([6.5, 6.0, 5.0, 13 / 3]) if n.endswith("gamma")
+            else np.array([-13.5, -13.8, -14.0, -14.5]))
+        for n in pta.params
+    }
+    nm = NMFp(psrs, pta.rn_containers)
+    freqs = np.linspace(3e-9, 5e-8, 12)
+    eng = FpEngine(psrs, Nvecs, Ts, device="cpu")
+    eng.precompute(freqs)
+    eng.enable_draw_compression(
+        [c.var_slice for c in pta.rn_containers],
+        [c.get_phiinv(noise) for c in pta.rn_containers],
+    )
+    assert all(blk.comp is not None for blk in eng.blocks)
+    from fastfp_amd.noise import batch_phiinv
+
+    piv = [p if p.dim() == 2 else p[None]
+           for p in batch_phiinv(pta.rn_containers, samples)]
+    margin = eng.compression_margin(piv)
+    assert 1.5 < margin < 1e3, f"test shape should sit in (1.5, 1e3): {margin}"
+    got = nm.sweep(freqs, samples, Nvecs, Ts, device="cpu", engine=eng)
+    # engine still compressed after the sweep (the old threshold
+    # disabled it here)
+    assert all(blk.comp is not None for blk in eng.blocks)
+    direct = nm.sweep(freqs, samples, Nvecs, Ts, device="cpu",
+                      compress=False)
+    np.testing.assert_allclose(got, direct, rtol=1e-7)
+
+
 def test_compression_margin_fallback():
     """Draws with absurdly large phi (phiinv near the jitter floor)
     must disable compression and still produce the exact direct

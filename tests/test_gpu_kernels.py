@@ -362,6 +362,50 @@ def test_engine_direct_stacked_gpu_matches_cpu():
     np.testing.assert_allclose(got, want, rtol=1e-8)
 
 
+def test_low_margin_draws_compressed_gpu_vs_direct():
+    """Uniform-prior tail draws (margin just above the 1.5 guard) on
+    the GPU compressed kernels vs the GPU direct path — the accuracy
+    that justifies the r02 margin-threshold change."""
+    from fastfp_amd import FpEngine, get_mats_nmfp, initialize_pta, \
+        make_synthetic_pta
+    from fastfp_amd.noise import batch_phiinv
+
+    psrs = make_synthetic_pta(npsr=3, ntoa=1500, ntm=10, seed=61)
+    noise = {"gw_gamma": 13.0 / 3.0, "gw_log10_A": float(np.log10(2e-15))}
+    for p in psrs:
+        noise[f"{p.name}_red_noise_gamma"] = 13.0 / 3.0
+        noise[f"{p.name}_red_noise_log10_A"] = -14.5
+    pta = initialize_pta(psrs, noise, inc_cp=True, rn_comps=20, gwb_comps=20)
+    TNTs, Nvecs, Ts = get_mats_nmfp(pta, noise)
+    for c in pta.rn_containers:
+        c.to(DEV)
+    D = 4
+    samples = {
+        n: (np.array([6.5, 6.0, 5.0, 13 / 3]) if n.endswith("gamma")
+            else np.array([-13.5, -13.8, -14.0, -14.5]))
+        for n in pta.params
+    }
+    freqs = np.linspace(3e-9, 5e-8, 40)
+    eng = FpEngine(psrs, Nvecs, Ts, device=DEV)
+    eng.precompute(freqs)
+    eng.enable_draw_compression(
+        [c.var_slice for c in pta.rn_containers],
+        [c.get_phiinv(noise).to(DEV) for c in pta.rn_containers],
+    )
+    assert all(blk.comp is not None for blk in eng.blocks)
+    piv = [p if p.dim() == 2 else p[None]
+           for p in batch_phiinv(pta.rn_containers, samples)]
+    margin = eng.compression_margin(piv)
+    assert margin > 1.5, margin
+    comp = eng.sweep(phiinvs=piv).cpu().numpy()
+    eng.disable_draw_compression()
+    eng._stack_direct()
+    direct = eng.sweep(phiinvs=piv).cpu().numpy()
+    scale = np.abs(direct).max()
+    assert np.isfinite(comp).all()
+    assert np.abs(comp - direct).max() / scale < 1e-6
+
+
 def test_direct_sweep_large_m_gpu():
     """DIRECT (uncompressed) GPU sweep at basis size 128 < m <= 256:
     the rocSOLVER-factored + diag_inv + register-resident trsm path
