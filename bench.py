@@ -106,6 +106,15 @@ def build_problem(args, device, rank):
         active[name] = pool[name][0].clone()
     # syncing check, done once here so the captured step never syncs
     pta._phi_homog = check_batch_homogeneous(pta.rn_containers)
+    # one-time validity guard on the WHOLE pool (outside the timed
+    # region): every draw must clear the compression margin the
+    # accuracy evidence covers (engine.compression_margin docstring)
+    if not args.no_compress and any(b.comp is not None for b in eng.blocks):
+        flat = {k: v.reshape(-1) for k, v in pool.items()}  # all rotations
+        piv = batch_phiinv(pta.rn_containers, flat, homogeneous=pta._phi_homog)
+        piv = [p if p.dim() == 2 else p[None] for p in piv]
+        margin = eng.compression_margin(piv)
+        assert margin > 1.5, f"draw pool margin {margin} below the guard"
     return pta, eng, pool, active
 
 
@@ -222,6 +231,11 @@ def main():
     value = evals / elapsed
     ms_per_step = elapsed / args.steps * 1000.0
 
+    # honesty guard (outside the timed region): the gathered spectrum
+    # must be finite everywhere
+    finite = bool(torch.isfinite(full).all().item())
+    assert finite, "non-finite Fp values in the benchmark spectrum"
+
     # secondary (untimed-region) measurement: the DIRECT full-m path
     # with the Schur draw compression off — reported alongside the
     # compressed headline so both numbers are on record.
@@ -287,6 +301,7 @@ def main():
                 "parallelism": f"dp{world} draw-sharded",
                 "spectrum_shape": list(full.shape),
                 "pool_rotations": R,
+                "spectrum_finite": finite,
             },
         }
         if direct is not None:
