@@ -204,6 +204,103 @@ extern "C" __global__ __launch_bounds__(512) void sbgemm_kernel(
 }
 
 // ---------------------------------------------------------------------
+// sbgemm_db: DOUBLE-BUFFERED variant of sbgemm — the next K-tile's T
+// panel + generated trig panel are staged while the current tile's
+// MFMAs run, so there is ONE barrier per K-tile instead of two and
+// the global loads + the ~100-cycle fp64 sincos hide under the MFMA
+// phase.  LDS doubles to ~50 KB (still 2+ workgroups/CU at the
+// 112-VGPR occupancy).  A/B arm: FASTFP_SBGEMM_ALGO=db.
+// ---------------------------------------------------------------------
+extern "C" __global__ __launch_bounds__(512) void sbgemm_db_kernel(
+    const double* __restrict__ T /*(ntoa, m) row-major*/,
+    const double* __restrict__ toas, const double* __restrict__ ninv,
+    const double* __restrict__ freqs, int ntoa, int m, int mp, int F2,
+    double* __restrict__ out, long plane_stride, long ldo) {
+  __shared__ double lT[2][16][FASTFP_MAXMP + 1];
+  __shared__ double lS[2][64][17];
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wv = tid >> 6;
+  const int c0 = blockIdx.x * 64;
+  const int mbase = blockIdx.z * FASTFP_MAXMP;
+  const int mp_loc = min(FASTFP_MAXMP, mp - mbase);
+  const int jw = (wv & 3) * 16;
+  const int rh = wv >> 2;
+  const int rowbase = rh * 64;
+  const int nrt_tot = mp_loc >> 4;
+  const int rt_lo = min(rh * 4, nrt_tot);
+  const int rt_hi = min(rt_lo + 4, nrt_tot);
+
+  const int ks = gridDim.y;
+  const int kchunk = (ntoa + ks - 1) / ks;
+  const int kbeg = blockIdx.y * kchunk;
+  const int kend = min(ntoa, kbeg + kchunk);
+  double* outp = out + (long)blockIdx.y * plane_stride;
+
+  f64x4 acc[4];
+#pragma unroll
+  for (int q = 0; q < 4; ++q) acc[q] = f64x4{0, 0, 0, 0};
+
+  // staging of K-tile k0 into buffer b
+#define SB_STAGE(k0, b)                                                    \
+  {                                                                        \
+    for (int idx = tid; idx < 16 * mp_loc; idx += 512) {                   \
+      const int k = idx / mp_loc, j = idx % mp_loc;                        \
+      const int gk = (k0) + k;                                             \
+      const int gj = mbase + j;                                            \
+      lT[b][k][j] = (gk < kend && gj < m) ? T[(long)gk * m + gj] : 0.0;    \
+    }                                                                      \
+    for (int idx = tid; idx < 32 * 16; idx += 512) {                       \
+      const int p = idx / 16, k = idx % 16;                                \
+      const int gk = (k0) + k;                                             \
+      const int ceven = c0 + 2 * p;                                        \
+      double sv = 0.0, cv = 0.0;                                           \
+      if (gk < kend && ceven < F2) {                                       \
+        const double wf = 2.0 * M_PI * freqs[ceven >> 1];                  \
+        sincos(wf * toas[gk], &sv, &cv);                                   \
+        const double ni = ninv[gk];                                        \
+        sv *= ni; cv *= ni;                                                \
+      }                                                                    \
+      lS[b][2 * p][k] = sv;                                                \
+      lS[b][2 * p + 1][k] = cv;                                            \
+    }                                                                      \
+  }
+
+  SB_STAGE(kbeg, 0);
+  __syncthreads();
+  int buf = 0;
+  for (int k0 = kbeg; k0 < kend; k0 += 16) {
+    if (k0 + 16 < kend) SB_STAGE(k0 + 16, buf ^ 1);
+#pragma unroll
+    for (int kk = 0; kk < 4; ++kk) {
+      const double b = lS[buf][jw + (lane & 15)][kk * 4 + (lane >> 4)];
+#pragma unroll
+      for (int q = 0; q < 4; ++q) {
+        if (rt_lo + q >= rt_hi) break;
+        const double a =
+            lT[buf][kk * 4 + (lane >> 4)][rowbase + q * 16 + (lane & 15)];
+        acc[q] = MFMA_F64(a, b, acc[q]);
+      }
+    }
+    buf ^= 1;
+    __syncthreads();
+  }
+#undef SB_STAGE
+
+#pragma unroll
+  for (int q = 0; q < 4; ++q) {
+    if (rt_lo + q >= rt_hi) break;
+#pragma unroll
+    for (int v = 0; v < 4; ++v) {
+      const int row = mbase + rowbase + q * 16 + 4 * v + (lane >> 4);
+      const int col = c0 + jw + (lane & 15);
+      if (col < F2) outp[(long)row * ldo + col] = acc[q][v];
+    }
+  }
+}
+
+// ---------------------------------------------------------------------
 // chol_batch: per draw d, assemble Sigma = TNT + diag(phiinv[d]) (+
 // identity padding to mp) in LDS, factor L L^T = Sigma (right-looking,
 // NB=16 blocks, MFMA trailing updates), invert each 16x16 diagonal
@@ -1084,6 +1181,14 @@ void launch_sbgemm(const double* T, const double* toas, const double* ninv,
                    hipStream_t stream) {
   const int ctiles = (F2 + 63) / 64;
   const int mtiles = (mp + FASTFP_MAXMP - 1) / FASTFP_MAXMP;
+  static const char* algo = getenv("FASTFP_SBGEMM_ALGO");
+  static const bool use_db = algo && 0 == __builtin_strcmp(algo, "db");
+  if (use_db) {
+    hipLaunchKernelGGL(sbgemm_db_kernel, dim3(ctiles, ksplit, mtiles),
+                       dim3(512), 0, stream, T, toas, ninv, freqs, ntoa, m,
+                       mp, F2, out, plane_stride, ldo);
+    return;
+  }
   hipLaunchKernelGGL(sbgemm_kernel, dim3(ctiles, ksplit, mtiles), dim3(512),
                      0, stream, T, toas, ninv, freqs, ntoa, m, mp, F2, out,
                      plane_stride, ldo);
