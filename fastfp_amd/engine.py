@@ -487,6 +487,29 @@ class FpEngine:
             return float("inf")
         return float(torch.stack(mins).min())
 
+    def compression_margin_per_draw(self, phiinvs) -> torch.Tensor:
+        """Per-draw compression margin: (D,) tensor of
+        min over pulsars/bins of phiinv_d / delta0.  Basis of the
+        per-draw hybrid split (NMFp.sweep): draws above the guard run
+        compressed, the (rare) prior-corner draws run the exact direct
+        path — one extreme draw no longer forces a whole batch off the
+        fast path.  Returns +inf per draw when nothing is compressed."""
+        mins = []
+        for blk, pinv in zip(self.blocks, phiinvs):
+            if blk.comp is None:
+                continue
+            p = _t64(pinv, self.device)
+            p = p[None, :] if p.dim() == 1 else p
+            mins.append(
+                (p[:, blk.comp["var"]] / blk.comp["delta0"][None, :])
+                .min(dim=1).values
+            )
+        if not mins:
+            D = phiinvs[0].shape[0] if phiinvs[0].dim() == 2 else 1
+            return torch.full((D,), float("inf"), dtype=torch.float64,
+                              device=self.device)
+        return torch.stack(mins).min(dim=0).values
+
     def disable_draw_compression(self):
         for blk in self.blocks:
             blk.comp = None
@@ -502,6 +525,7 @@ class FpEngine:
         sigmas=None,
         draw_chunk: int = 32,
         accumulate_to=None,
+        force_direct: bool = False,
     ) -> torch.Tensor:
         """Run the Fp sweep over the precomputed frequency grid.
 
@@ -509,6 +533,10 @@ class FpEngine:
         Fp path or (D, m) for D noise draws.  Alternatively pass dense
         ``sigmas`` (m, m) / (D, m, m) directly (the ``get_mats_fp``
         contract).  Returns Fp of shape (F,) or (D, F).
+
+        ``force_direct``: bypass the Schur compression for THIS call
+        (used by the per-draw hybrid for prior-corner draws whose
+        margin is below the guard).
         """
         assert self.freqs is not None, "call precompute(freqs) first"
         F = self.freqs.shape[0]
@@ -526,13 +554,14 @@ class FpEngine:
             fp = torch.zeros((D, F), dtype=torch.float64, device=self.device)
 
         stack = getattr(self, "_comp_stack", None)
-        if stack is not None and phiinvs is not None:
+        if stack is not None and phiinvs is not None and not force_direct:
             return self._sweep_stacked(phiinvs, fp, D, F, draw_chunk, batched)
         dstack = getattr(self, "_direct_stack", None)
         if (
             dstack is not None
             and phiinvs is not None
-            and all(blk.comp is None for blk in self.blocks)
+            and (force_direct
+                 or all(blk.comp is None for blk in self.blocks))
         ):
             return self._sweep_stacked_direct(
                 phiinvs, fp, D, F, draw_chunk, batched
@@ -570,7 +599,7 @@ class FpEngine:
                             torch.diagonal(sigma, dim1=-2, dim2=-1)
                             - torch.diagonal(blk.TNT)[None, :]
                         )
-                    if blk.comp is not None:
+                    if blk.comp is not None and not force_direct:
                         # Schur-compressed: C_d = diag(1/Delta_d) + G
                         c = blk.comp
                         phi_var = (

@@ -113,7 +113,7 @@ class NMFp:
                                homogeneous=self._phi_homog)
         # scalar-parameter dicts produce (m,) vectors; promote to (1, m)
         phiinvs = [p[None, :] if p.dim() == 1 else p for p in phiinvs]
-        # Margin guard threshold: the compressed path is ALGEBRAICALLY
+        # Per-draw margin guard: the compressed path is ALGEBRAICALLY
         # exact for any Delta_d = phiinv_d - delta0 > 0; the only
         # numerical hazard is the subtraction's precision loss as
         # margin -> 1 (rel err = eps/(1 - 1/margin)) and the Delta <= 0
@@ -122,19 +122,37 @@ class NMFp:
         # 2e-10 of the spectrum scale (docs/PERFORMANCE.md r02) — the
         # round-1 threshold of 1e3 was ~3 orders too conservative and
         # silently pushed whole uniform-prior sweeps onto the 4x-slower
-        # direct path.
-        if compress and engine.compression_margin(phiinvs) < 1.5:
-            # draws at/below the Sigma_0 jitter floor:
-            # fall back to the exact direct path
-            if engine._use_hip and any(b.m > 256 for b in engine.blocks):
-                raise RuntimeError(
-                    "noise draws imply red-noise phi so large that the "
-                    "compressed solve is numerically unsafe, and the "
-                    "direct GPU solve does not support basis size m > "
-                    "256; run these draws on the CPU engine "
-                    "(device='cpu') or restrict the prior range"
-                )
-            engine.disable_draw_compression()
+        # direct path.  The guard is PER DRAW: prior-corner draws
+        # (margin < 1.5) run the exact direct path, the rest stay
+        # compressed — one extreme draw no longer slows a whole batch.
+        if compress and any(b.comp is not None for b in engine.blocks):
+            margins = engine.compression_margin_per_draw(phiinvs)
+            bad = margins < 1.5
+            nbad = int(bad.sum())
+            if nbad:
+                if engine._use_hip and any(b.m > 256 for b in engine.blocks):
+                    raise RuntimeError(
+                        "noise draws imply red-noise phi so large that "
+                        "the compressed solve is numerically unsafe, and "
+                        "the direct GPU solve does not support basis "
+                        "size m > 256; run these draws on the CPU engine "
+                        "(device='cpu') or restrict the prior range"
+                    )
+                good_idx = torch.nonzero(~bad).reshape(-1)
+                bad_idx = torch.nonzero(bad).reshape(-1)
+                D = margins.shape[0]
+                F = engine.freqs.shape[0]
+                fp = torch.zeros((D, F), dtype=torch.float64,
+                                 device=engine.device)
+                if len(good_idx):
+                    sub = [p[good_idx.to(p.device)] for p in phiinvs]
+                    fp[good_idx] = engine.sweep(phiinvs=sub,
+                                                draw_chunk=draw_chunk)
+                sub = [p[bad_idx.to(p.device)] for p in phiinvs]
+                fp[bad_idx] = engine.sweep(phiinvs=sub,
+                                           draw_chunk=draw_chunk,
+                                           force_direct=True)
+                return fp.cpu().numpy()
         if os.environ.get("FASTFP_CLI_TIMING") == "1":
             import time
 

@@ -274,6 +274,47 @@ def test_compression_low_margin_draws_stay_compressed_and_exact():
     np.testing.assert_allclose(got, direct, rtol=1e-7)
 
 
+def test_per_draw_hybrid_split_matches_direct():
+    """A batch mixing normal draws with ONE prior-corner draw (margin
+    below the guard): the corner draw runs the exact direct path, the
+    rest stay compressed, and the combined output equals the all-direct
+    reference."""
+    psrs = make_synthetic_pta(npsr=2, ntoa=400, ntm=8, seed=17)
+    noise = {"gw_gamma": 13.0 / 3.0, "gw_log10_A": float(np.log10(2e-15))}
+    for p in psrs:
+        noise[f"{p.name}_red_noise_gamma"] = 13.0 / 3.0
+        noise[f"{p.name}_red_noise_log10_A"] = -14.5
+    pta = initialize_pta(psrs, noise, inc_cp=True, rn_comps=10, gwb_comps=10)
+    TNTs, Nvecs, Ts = get_mats_nmfp(pta, noise)
+    samples = {
+        n: (np.array([4.0, 9.0, 13 / 3]) if n.endswith("gamma")
+            else np.array([-14.5, -12.5, -15.0]))
+        for n in pta.params
+    }
+    nm = NMFp(psrs, pta.rn_containers)
+    freqs = np.linspace(3e-9, 5e-8, 8)
+    eng = FpEngine(psrs, Nvecs, Ts, device="cpu")
+    eng.precompute(freqs)
+    eng.enable_draw_compression(
+        [c.var_slice for c in pta.rn_containers],
+        [c.get_phiinv(noise) for c in pta.rn_containers],
+    )
+    assert all(blk.comp is not None for blk in eng.blocks)
+    from fastfp_amd.noise import batch_phiinv
+
+    piv = [p if p.dim() == 2 else p[None]
+           for p in batch_phiinv(pta.rn_containers, samples)]
+    margins = eng.compression_margin_per_draw(piv).numpy()
+    assert margins[1] < 1.5 <= min(margins[0], margins[2]), margins
+    got = nm.sweep(freqs, samples, Nvecs, Ts, device="cpu", engine=eng)
+    # compression must survive the sweep (the old guard disabled it)
+    assert all(blk.comp is not None for blk in eng.blocks)
+    direct = nm.sweep(freqs, samples, Nvecs, Ts, device="cpu",
+                      compress=False)
+    assert np.isfinite(got).all()
+    np.testing.assert_allclose(got, direct, rtol=1e-7)
+
+
 def test_compression_margin_fallback():
     """Draws with absurdly large phi (phiinv near the jitter floor)
     must disable compression and still produce the exact direct
