@@ -314,9 +314,20 @@ extern "C" __global__ __launch_bounds__(512) void sbgemm_db_kernel(
 // shave Ash from 33.3 to 32.0 KB at mp=64 — one more workgroup per CU
 // — but the extra address arithmetic measured +16 VGPR, dropping a
 // waves/SIMD tier: net zero.  Padded layout kept.)
-#define A_(i, j) Ash[(i) * (NBT * 16 + 1) + (j)]
+//
+// TRI (FASTFP_CHOL_TRI=1, NBT <= 4): LOWER-TRIANGLE-packed Sigma —
+// row i holds i+1 entries (+1 pad), halving LDS (33.3 -> 17.4 KB at
+// mp=64) so ~7 workgroups fit per CU instead of 4; concurrency is the
+// suspected chol bottleneck (82% parked waves resist every other
+// explanation tested).  All factor-phase accesses are at-or-below the
+// diagonal; the diagonal tile's upper half is read through the
+// symmetric mirror AS_, and trailing stores into diagonal tiles are
+// lower-guarded.
+#define A_(i, j)                                                        \
+  Ash[TRI ? ((i) * ((i) + 3) / 2 + (j)) : ((i) * (NBT * 16 + 1) + (j))]
+#define AS_(i, j) ((j) <= (i) ? A_(i, j) : A_(j, i))
 
-template <int NBT>
+template <int NBT, bool TRI = false>
 __global__ __launch_bounds__(512, 4) void chol_batch_kernel(
     const double* __restrict__ TNT_all /*(P,m,m)*/,
     const double* __restrict__ phiinv_all /*(P,D,m)*/, int m, int D,
@@ -333,7 +344,7 @@ __global__ __launch_bounds__(512, 4) void chol_batch_kernel(
   // serial diagonal phases then overlap ACROSS workgroups -- the PMC
   // profile showed 84% of wave cycles parked, profiles/).
   constexpr int mp = NBT * 16;
-  __shared__ double Ash[mp * (mp + 1)];
+  __shared__ double Ash[TRI ? mp * (mp + 3) / 2 : mp * (mp + 1)];
   __shared__ double inv16[2][16][17];
 
   const int d = blockIdx.x;
@@ -345,9 +356,10 @@ __global__ __launch_bounds__(512, 4) void chol_batch_kernel(
 
   const int nthr = blockDim.x;
   const int nwv = nthr >> 6;
-  // assemble Sigma in LDS
+  // assemble Sigma in LDS (lower triangle only under TRI)
   for (int idx = tid; idx < mp * mp; idx += nthr) {
     const int i = idx / mp, j = idx % mp;
+    if (TRI && j > i) continue;
     double v = (i < m && j < m) ? TNT[(long)i * m + j] : 0.0;
     if (i == j) v += (i < m) ? phiinv[(long)d * m + i] : 1.0;
     A_(i, j) = v;
@@ -366,7 +378,7 @@ __global__ __launch_bounds__(512, 4) void chol_batch_kernel(
     const int i = lane & 15;                                                 \
     double row[16];                                                          \
     _Pragma("unroll") for (int c = 0; c < 16; ++c) row[c] =                  \
-        A_(dk0 + i, dk0 + c);                                                \
+        AS_(dk0 + i, dk0 + c);                                               \
     _Pragma("unroll") for (int t = 0; t < 16; ++t) {                         \
       const double dv = sqrt(__shfl(row[t], t, 16));                         \
       const double rdv = 1.0 / dv;                                           \
@@ -434,9 +446,11 @@ __global__ __launch_bounds__(512, 4) void chol_batch_kernel(
     for (int ib = kb + 1 + wv; ib < nb; ib += nwv) {
       const int i0 = ib * NB;
       f64x4 uacc;
+      // diagonal tile (i0 == j1): upper half read via the symmetric
+      // mirror under TRI, and its stores lower-guarded
 #pragma unroll
       for (int v = 0; v < 4; ++v)
-        uacc[v] = A_(i0 + 4 * v + (lane >> 4), j1 + (lane & 15));
+        uacc[v] = AS_(i0 + 4 * v + (lane >> 4), j1 + (lane & 15));
 #pragma unroll
       for (int kk = 0; kk < 4; ++kk) {
         const double a = -A_(i0 + (lane & 15), k0 + kk * 4 + (lane >> 4));
@@ -445,7 +459,8 @@ __global__ __launch_bounds__(512, 4) void chol_batch_kernel(
       }
 #pragma unroll
       for (int v = 0; v < 4; ++v)
-        A_(i0 + 4 * v + (lane >> 4), j1 + (lane & 15)) = uacc[v];
+        if (!TRI || i0 + 4 * v + (lane >> 4) >= j1 + (lane & 15))
+          A_(i0 + 4 * v + (lane >> 4), j1 + (lane & 15)) = uacc[v];
     }
     __syncthreads();
 
@@ -467,7 +482,7 @@ __global__ __launch_bounds__(512, 4) void chol_batch_kernel(
         f64x4 uacc;
 #pragma unroll
         for (int v = 0; v < 4; ++v)
-          uacc[v] = A_(i0 + 4 * v + (lane >> 4), j0 + (lane & 15));
+          uacc[v] = AS_(i0 + 4 * v + (lane >> 4), j0 + (lane & 15));
 #pragma unroll
         for (int kk = 0; kk < 4; ++kk) {
           const double a = -A_(i0 + (lane & 15), k0 + kk * 4 + (lane >> 4));
@@ -476,16 +491,18 @@ __global__ __launch_bounds__(512, 4) void chol_batch_kernel(
         }
 #pragma unroll
         for (int v = 0; v < 4; ++v)
-          A_(i0 + 4 * v + (lane >> 4), j0 + (lane & 15)) = uacc[v];
+          if (!TRI || i0 + 4 * v + (lane >> 4) >= j0 + (lane & 15))
+            A_(i0 + 4 * v + (lane >> 4), j0 + (lane & 15)) = uacc[v];
       }
     }
     __syncthreads();
   }
 
-  // write back L (full rows; upper-triangle junk is never read)
+  // write back L (full rows; upper-triangle junk is never read — TRI
+  // writes deterministic zeros there)
   for (int idx = tid; idx < mp * mp; idx += nthr) {
     const int i = idx / mp, j = idx % mp;
-    L[((long)d * mp + i) * mp + j] = A_(i, j);
+    L[((long)d * mp + i) * mp + j] = (TRI && j > i) ? 0.0 : A_(i, j);
   }
 }
 
@@ -1200,6 +1217,19 @@ void launch_chol_batch(const double* TNT, const double* phiinv, int m, int mp,
   // small matrices: 256 threads -> 4 workgroups/CU despite the 115-VGPR
   // diagonal-factor pressure; large: 512 threads for MFMA coverage
   const dim3 grid(D, P), blk(mp <= 64 ? 256 : 512);
+  // FASTFP_CHOL_TRI=1: lower-triangle-packed Sigma (A/B arm, mp<=64)
+  static const char* tri_env = getenv("FASTFP_CHOL_TRI");
+  static const bool tri = tri_env && tri_env[0] == '1';
+  if (tri && mp <= 64) {
+    switch (mp >> 4) {
+#define CHOL_TRI_CASE(NBT) \
+      case NBT: hipLaunchKernelGGL((chol_batch_kernel<NBT, true>), grid, \
+                    blk, 0, stream, TNT, phiinv, m, D, L, invd); break;
+      CHOL_TRI_CASE(1) CHOL_TRI_CASE(2) CHOL_TRI_CASE(3) CHOL_TRI_CASE(4)
+#undef CHOL_TRI_CASE
+    }
+    return;
+  }
   switch (mp >> 4) {
 #define CHOL_CASE(NBT) \
     case NBT: hipLaunchKernelGGL(chol_batch_kernel<NBT>, grid, blk, 0, \
