@@ -690,7 +690,13 @@ __global__ __launch_bounds__(512, 4) void trsm_fp_kernel(
 #pragma unroll
   for (int e = 0; e < DPG; ++e) {
     if (e >= ndr) break;
-    double pss = 0, pcc = 0, psc = 0, psu = 0, pcu = 0;
+    // lane-split accumulation: a frequency's sin column is an even
+    // lane, its cos column the adjacent odd lane, and the quadratics
+    // pair up — w*w accumulates ss on even lanes and cc on odd lanes
+    // in the SAME register (likewise w*wu -> su/cu; w*wp is sc on
+    // both).  3 fma chains instead of 5; the odd-lane halves arrive
+    // by two shfl_xor after the row-group reduction.
+    double a1 = 0, a2 = 0, a3 = 0;
 #pragma unroll
     for (int rt = 0; rt < NBT; ++rt) {
 #pragma unroll
@@ -698,28 +704,25 @@ __global__ __launch_bounds__(512, 4) void trsm_fp_kernel(
         const double w = W[e][rt][v];
         const double wp = __shfl_xor(w, 1, 64);  // partner column
         const double wu = Wu[e][rt * 16 + 4 * v + lk];
-        // even lanes own the (sin, cos) pair of their frequency
-        pss = fma(w, w, pss);
-        pcc = fma(wp, wp, pcc);
-        psc = fma(w, wp, psc);
-        psu = fma(w, wu, psu);
-        pcu = fma(wp, wu, pcu);
+        a1 = fma(w, w, a1);
+        a2 = fma(w, wu, a2);
+        a3 = fma(w, wp, a3);
       }
     }
     // sum the 4 row groups (lanes lk = 0..3 share li): +16, +32 lanes
-    pss += __shfl_down(pss, 32, 64); pss += __shfl_down(pss, 16, 64);
-    pcc += __shfl_down(pcc, 32, 64); pcc += __shfl_down(pcc, 16, 64);
-    psc += __shfl_down(psc, 32, 64); psc += __shfl_down(psc, 16, 64);
-    psu += __shfl_down(psu, 32, 64); psu += __shfl_down(psu, 16, 64);
-    pcu += __shfl_down(pcu, 32, 64); pcu += __shfl_down(pcu, 16, 64);
+    a1 += __shfl_down(a1, 32, 64); a1 += __shfl_down(a1, 16, 64);
+    a2 += __shfl_down(a2, 32, 64); a2 += __shfl_down(a2, 16, 64);
+    a3 += __shfl_down(a3, 32, 64); a3 += __shfl_down(a3, 16, 64);
+    const double b1 = __shfl_xor(a1, 1, 64);  // cc (for even lanes)
+    const double b2 = __shfl_xor(a2, 1, 64);  // cu
     const int q = (jw + li) >> 1;  // frequency index within the block
     if (lk == 0 && (li & 1) == 0 && q < FPT_FREQS && f0 + q < F) {
       const int f = f0 + q;
-      const double M11 = sNs[f] - gsign * pss;
-      const double M22 = sNs[F + f] - gsign * pcc;
-      const double M12 = sNs[2 * F + f] - gsign * psc;
-      const double N1 = sNr[f] - gsign * psu;
-      const double N2 = sNr[F + f] - gsign * pcu;
+      const double M11 = sNs[f] - gsign * a1;
+      const double M22 = sNs[F + f] - gsign * b1;
+      const double M12 = sNs[2 * F + f] - gsign * a3;
+      const double N1 = sNr[f] - gsign * a2;
+      const double N2 = sNr[F + f] - gsign * b2;
       const double det = fma(M11, M22, -M12 * M12);
       const double num =
           fma(N1 * N1, M22, fma(-2.0 * N1, N2 * M12, N2 * N2 * M11));
@@ -853,7 +856,9 @@ __global__ __launch_bounds__(512, 4) void trsm_fp_rl_kernel(
   }
   __syncthreads();
 
-  double pss = 0, pcc = 0, psc = 0, psu = 0, pcu = 0;
+  // lane-split accumulation (see trsm_fp_kernel): 3 fma chains, odd
+  // halves fetched by shfl_xor after the row-group reduction
+  double a1 = 0, a2 = 0, a3 = 0;
 #pragma unroll
   for (int rt = 0; rt < NBT; ++rt) {
 #pragma unroll
@@ -861,26 +866,24 @@ __global__ __launch_bounds__(512, 4) void trsm_fp_rl_kernel(
       const double w = W[rt][v];
       const double wp = __shfl_xor(w, 1, 64);
       const double wu = Wu[rt * 16 + 4 * v + lk];
-      pss = fma(w, w, pss);
-      pcc = fma(wp, wp, pcc);
-      psc = fma(w, wp, psc);
-      psu = fma(w, wu, psu);
-      pcu = fma(wp, wu, pcu);
+      a1 = fma(w, w, a1);
+      a2 = fma(w, wu, a2);
+      a3 = fma(w, wp, a3);
     }
   }
-  pss += __shfl_down(pss, 32, 64); pss += __shfl_down(pss, 16, 64);
-  pcc += __shfl_down(pcc, 32, 64); pcc += __shfl_down(pcc, 16, 64);
-  psc += __shfl_down(psc, 32, 64); psc += __shfl_down(psc, 16, 64);
-  psu += __shfl_down(psu, 32, 64); psu += __shfl_down(psu, 16, 64);
-  pcu += __shfl_down(pcu, 32, 64); pcu += __shfl_down(pcu, 16, 64);
+  a1 += __shfl_down(a1, 32, 64); a1 += __shfl_down(a1, 16, 64);
+  a2 += __shfl_down(a2, 32, 64); a2 += __shfl_down(a2, 16, 64);
+  a3 += __shfl_down(a3, 32, 64); a3 += __shfl_down(a3, 16, 64);
+  const double b1 = __shfl_xor(a1, 1, 64);
+  const double b2 = __shfl_xor(a2, 1, 64);
   const int q = (jw + li) >> 1;
   if (lk == 0 && (li & 1) == 0 && q < FPT_FREQS && f0 + q < F) {
     const int f = f0 + q;
-    const double M11 = sNs[f] - gsign * pss;
-    const double M22 = sNs[F + f] - gsign * pcc;
-    const double M12 = sNs[2 * F + f] - gsign * psc;
-    const double N1 = sNr[f] - gsign * psu;
-    const double N2 = sNr[F + f] - gsign * pcu;
+    const double M11 = sNs[f] - gsign * a1;
+    const double M22 = sNs[F + f] - gsign * b1;
+    const double M12 = sNs[2 * F + f] - gsign * a3;
+    const double N1 = sNr[f] - gsign * a2;
+    const double N2 = sNr[F + f] - gsign * b2;
     const double det = fma(M11, M22, -M12 * M12);
     const double num =
         fma(N1 * N1, M22, fma(-2.0 * N1, N2 * M12, N2 * N2 * M11));
@@ -1015,7 +1018,9 @@ __global__ __launch_bounds__(512, 4) void trsm_fp_res_kernel(
   }
   __syncthreads();
 
-  double pss = 0, pcc = 0, psc = 0, psu = 0, pcu = 0;
+  // lane-split accumulation (see trsm_fp_kernel): 3 fma chains, odd
+  // halves fetched by shfl_xor after the row-group reduction
+  double a1 = 0, a2 = 0, a3 = 0;
 #pragma unroll
   for (int rt = 0; rt < NBT; ++rt) {
 #pragma unroll
@@ -1023,26 +1028,24 @@ __global__ __launch_bounds__(512, 4) void trsm_fp_res_kernel(
       const double w = W[rt][v];
       const double wp = __shfl_xor(w, 1, 64);
       const double wu = Wu[rt * 16 + 4 * v + lk];
-      pss = fma(w, w, pss);
-      pcc = fma(wp, wp, pcc);
-      psc = fma(w, wp, psc);
-      psu = fma(w, wu, psu);
-      pcu = fma(wp, wu, pcu);
+      a1 = fma(w, w, a1);
+      a2 = fma(w, wu, a2);
+      a3 = fma(w, wp, a3);
     }
   }
-  pss += __shfl_down(pss, 32, 64); pss += __shfl_down(pss, 16, 64);
-  pcc += __shfl_down(pcc, 32, 64); pcc += __shfl_down(pcc, 16, 64);
-  psc += __shfl_down(psc, 32, 64); psc += __shfl_down(psc, 16, 64);
-  psu += __shfl_down(psu, 32, 64); psu += __shfl_down(psu, 16, 64);
-  pcu += __shfl_down(pcu, 32, 64); pcu += __shfl_down(pcu, 16, 64);
+  a1 += __shfl_down(a1, 32, 64); a1 += __shfl_down(a1, 16, 64);
+  a2 += __shfl_down(a2, 32, 64); a2 += __shfl_down(a2, 16, 64);
+  a3 += __shfl_down(a3, 32, 64); a3 += __shfl_down(a3, 16, 64);
+  const double b1 = __shfl_xor(a1, 1, 64);
+  const double b2 = __shfl_xor(a2, 1, 64);
   const int q = (jw + li) >> 1;
   if (lk == 0 && (li & 1) == 0 && q < FPT_FREQS && f0 + q < F) {
     const int f = f0 + q;
-    const double M11 = sNs[f] - gsign * pss;
-    const double M22 = sNs[F + f] - gsign * pcc;
-    const double M12 = sNs[2 * F + f] - gsign * psc;
-    const double N1 = sNr[f] - gsign * psu;
-    const double N2 = sNr[F + f] - gsign * pcu;
+    const double M11 = sNs[f] - gsign * a1;
+    const double M22 = sNs[F + f] - gsign * b1;
+    const double M12 = sNs[2 * F + f] - gsign * a3;
+    const double N1 = sNr[f] - gsign * a2;
+    const double N2 = sNr[F + f] - gsign * b2;
     const double det = fma(M11, M22, -M12 * M12);
     const double num =
         fma(N1 * N1, M22, fma(-2.0 * N1, N2 * M12, N2 * N2 * M11));
