@@ -327,6 +327,9 @@ extern "C" __global__ __launch_bounds__(512) void sbgemm_db_kernel(
   Ash[TRI ? ((i) * ((i) + 3) / 2 + (j)) : ((i) * (NBT * 16 + 1) + (j))]
 #define AS_(i, j) ((j) <= (i) ? A_(i, j) : A_(j, i))
 
+// (Forcing a 6-waves/SIMD allocation target on the small shapes got
+// there only by spilling 44 B/lane and measured neutral — reverted;
+// the no-spill allocation already reaches 5 waves with TRI.)
 template <int NBT, bool TRI = false>
 __global__ __launch_bounds__(512, 4) void chol_batch_kernel(
     const double* __restrict__ TNT_all /*(P,m,m)*/,
