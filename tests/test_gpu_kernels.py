@@ -454,6 +454,55 @@ def test_direct_sweep_large_m_gpu():
     np.testing.assert_allclose(gpu, cpu, rtol=1e-7)
 
 
+def test_graphed_nmfp_sweep_matches_eager(monkeypatch):
+    """The CLI's hipGraph-replayed batch sweep (NMFp._sweep_graphed)
+    must reproduce the eager sweep bitwise, including across repeated
+    replays with different draw batches, and must recompute
+    prior-corner draws through the direct path."""
+    from fastfp_amd import FpEngine, get_mats_nmfp, initialize_pta, \
+        make_synthetic_pta
+    from fastfp_amd.nmfp import NMFp
+
+    psrs = make_synthetic_pta(npsr=3, ntoa=800, ntm=8, seed=71)
+    noise = {"gw_gamma": 13.0 / 3.0, "gw_log10_A": float(np.log10(2e-15))}
+    for p in psrs:
+        noise[f"{p.name}_red_noise_gamma"] = 13.0 / 3.0
+        noise[f"{p.name}_red_noise_log10_A"] = -14.5
+    pta = initialize_pta(psrs, noise, inc_cp=True, rn_comps=12, gwb_comps=12)
+    TNTs, Nvecs, Ts = get_mats_nmfp(pta, noise)
+    for c in pta.rn_containers:
+        c.to(DEV)
+    eng = FpEngine(psrs, Nvecs, Ts, device=DEV)
+    freqs = np.linspace(3e-9, 5e-8, 30)
+    eng.precompute(freqs)
+    eng.enable_draw_compression(
+        [c.var_slice for c in pta.rn_containers],
+        [c.get_phiinv(noise).to(DEV) for c in pta.rn_containers],
+    )
+    assert eng._comp_stack is not None
+    nm = NMFp(psrs, pta.rn_containers)
+    rng = np.random.default_rng(9)
+    D = 16
+    batches = []
+    for _ in range(3):
+        batches.append({
+            n: (rng.uniform(2, 6, D) if n.endswith("gamma")
+                else rng.uniform(-16, -14, D))
+            for n in pta.params
+        })
+    # one batch includes a prior-corner draw (direct-path overwrite)
+    for n in pta.params:
+        batches[2][n][3] = 9.0 if n.endswith("gamma") else -12.5
+
+    monkeypatch.setenv("FASTFP_NO_GRAPH", "1")
+    eager = [nm.sweep(freqs, b, Nvecs, Ts, engine=eng) for b in batches]
+    monkeypatch.delenv("FASTFP_NO_GRAPH")
+    graphed = [nm.sweep(freqs, b, Nvecs, Ts, engine=eng) for b in batches]
+    assert nm._graphs, "graph path should have engaged"
+    for a, b in zip(eager, graphed):
+        np.testing.assert_array_equal(a, b)
+
+
 def test_graph_captured_sweep_bitwise_equals_eager():
     """A hipGraph-captured sweep (the bench's step structure) must
     reproduce the eager sweep BITWISE on replay."""
