@@ -1223,18 +1223,20 @@ void launch_chol_batch(const double* TNT, const double* phiinv, int m, int mp,
   // small matrices: 256 threads -> 4 workgroups/CU despite the 115-VGPR
   // diagonal-factor pressure; large: 512 threads for MFMA coverage
   const dim3 grid(D, P), blk(mp <= 64 ? 256 : 512);
-  // Lower-triangle-packed Sigma is the DEFAULT for mp <= 64 (measured
-  // -13% on chol at the bench shape: 3.50 -> 3.03 ms, the extra
-  // workgroup of concurrency); FASTFP_CHOL_TRI=0 restores the square
-  // layout for A/B.
+  // Lower-triangle-packed Sigma is the DEFAULT (measured -13% on chol
+  // at the bench shape from the extra workgroup of concurrency; at
+  // mp=128 the square layout is 132 KB LDS = ONE workgroup per CU, so
+  // packing doubles concurrency).  FASTFP_CHOL_TRI=0 restores the
+  // square layout for A/B.
   static const char* tri_env = getenv("FASTFP_CHOL_TRI");
   static const bool tri = !(tri_env && tri_env[0] == '0');
-  if (tri && mp <= 64) {
+  if (tri) {
     switch (mp >> 4) {
 #define CHOL_TRI_CASE(NBT) \
       case NBT: hipLaunchKernelGGL((chol_batch_kernel<NBT, true>), grid, \
                     blk, 0, stream, TNT, phiinv, m, D, L, invd); break;
       CHOL_TRI_CASE(1) CHOL_TRI_CASE(2) CHOL_TRI_CASE(3) CHOL_TRI_CASE(4)
+      CHOL_TRI_CASE(5) CHOL_TRI_CASE(6) CHOL_TRI_CASE(7) CHOL_TRI_CASE(8)
 #undef CHOL_TRI_CASE
     }
     return;
