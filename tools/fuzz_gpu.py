@@ -27,9 +27,10 @@ def one_case(seed):
     npsr = int(rng.integers(1, 6))
     ntoa = int(rng.integers(40, 4000))
     ntm = int(rng.integers(3, 10))
-    # up to 40 components: covers the compressed solve at NBT=5 (the
-    # substitution-kernel dispatch) as well as the small-NBT DPG path
-    rn = int(rng.integers(1, 41))
+    # up to 60 components: covers the compressed solve across the
+    # NBT template range AND (with gp_ecorr) direct solves beyond
+    # m=128 (the rocSOLVER-factored + diag_inv + right-looking path)
+    rn = int(rng.integers(1, 61))
     F = int(rng.integers(1, 300))
     D = int(rng.integers(1, 40))
     inc_cp = bool(rng.integers(0, 2))
@@ -59,6 +60,11 @@ def one_case(seed):
             else rng.uniform(-16, -14, D))
         for n in pta.params
     }
+    if D >= 3 and bool(rng.integers(0, 2)):
+        # inject prior-corner draws: exercises the per-draw hybrid
+        # (compressed batch + direct rows) on both devices
+        for n in pta.params:
+            samples[n][1] = 8.5 if n.endswith("gamma") else -12.6
     nm = NMFp(psrs, pta.rn_containers)
     cpu = nm.sweep(freqs, samples, Nvecs, Ts, device="cpu")
     for c in pta.rn_containers:
