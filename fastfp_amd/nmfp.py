@@ -121,7 +121,29 @@ class NMFp:
         return fstat
 
     # ------------------------------------------------------------------
-    def _sweep_graphed(self, engine, samples, draw_chunk, compress):
+    def _direct_rows(self, engine, phiinvs, bad_idx, draw_chunk, Nvecs, Ts):
+        """Exact direct-path evaluation of the draws at ``bad_idx``.
+
+        On GPU with basis size m <= 256 this is the direct HIP solve;
+        above 256 (no GPU kernel) the rows run through a cached CPU
+        LAPACK eager engine — slow but exact, and only the rare
+        prior-corner draws pay it."""
+        sub = [p[bad_idx.to(p.device)] for p in phiinvs]
+        if engine._use_hip and any(b.m > 256 for b in engine.blocks):
+            fr = engine.freqs.cpu().numpy()
+            ce = getattr(self, "_cpu_eng", None)
+            if ce is None or len(ce[0]) != len(fr) or not np.array_equal(ce[0], fr):
+                eng = FpEngine(self.psrs, Nvecs, Ts, device="cpu")
+                eng.precompute(fr)
+                self._cpu_eng = ce = (fr, eng)
+            vals = ce[1].sweep(phiinvs=[p.cpu() for p in sub],
+                               draw_chunk=draw_chunk)
+            return vals.to(engine.device)
+        return engine.sweep(phiinvs=sub, draw_chunk=draw_chunk,
+                            force_direct=True)
+
+    def _sweep_graphed(self, engine, samples, draw_chunk, compress,
+                       Nvecs, Ts):
         """hipGraph-replayed batch sweep, or ``None`` when any
         precondition fails (CPU engine, heterogeneous containers, no
         stacked compression, device-tensor samples, odd shapes) — the
@@ -168,10 +190,9 @@ class NMFp:
         bad = margins < 1.5
         if bool(bad.any()):
             bad_idx = torch.nonzero(bad).reshape(-1)
-            sub = [p[bad_idx.to(p.device)] for p in phiinvs]
             fp = fp.clone()
-            fp[bad_idx] = engine.sweep(phiinvs=sub, draw_chunk=draw_chunk,
-                                       force_direct=True)
+            fp[bad_idx] = self._direct_rows(engine, phiinvs, bad_idx,
+                                            draw_chunk, Nvecs, Ts)
         return fp.cpu().numpy()
 
     # ------------------------------------------------------------------
@@ -219,7 +240,8 @@ class NMFp:
         # CLI hot loop): phi assembly + the stacked compressed sweep
         # are captured once and replayed per batch — the per-batch
         # eager launch overhead was the remaining CLI-vs-bench gap.
-        out = self._sweep_graphed(engine, samples, draw_chunk, compress)
+        out = self._sweep_graphed(engine, samples, draw_chunk, compress,
+                                  Nvecs, Ts)
         if out is not None:
             return out
 
@@ -244,14 +266,6 @@ class NMFp:
             bad = margins < 1.5
             nbad = int(bad.sum())
             if nbad:
-                if engine._use_hip and any(b.m > 256 for b in engine.blocks):
-                    raise RuntimeError(
-                        "noise draws imply red-noise phi so large that "
-                        "the compressed solve is numerically unsafe, and "
-                        "the direct GPU solve does not support basis "
-                        "size m > 256; run these draws on the CPU engine "
-                        "(device='cpu') or restrict the prior range"
-                    )
                 good_idx = torch.nonzero(~bad).reshape(-1)
                 bad_idx = torch.nonzero(bad).reshape(-1)
                 D = margins.shape[0]
@@ -262,10 +276,8 @@ class NMFp:
                     sub = [p[good_idx.to(p.device)] for p in phiinvs]
                     fp[good_idx] = engine.sweep(phiinvs=sub,
                                                 draw_chunk=draw_chunk)
-                sub = [p[bad_idx.to(p.device)] for p in phiinvs]
-                fp[bad_idx] = engine.sweep(phiinvs=sub,
-                                           draw_chunk=draw_chunk,
-                                           force_direct=True)
+                fp[bad_idx] = self._direct_rows(engine, phiinvs, bad_idx,
+                                                draw_chunk, Nvecs, Ts)
                 return fp.cpu().numpy()
         if os.environ.get("FASTFP_CLI_TIMING") == "1":
             import time
