@@ -35,3 +35,27 @@ def test_bench_json_contract():
 def test_bench_presets_parse():
     j = _run("--preset", "ecorr67")
     assert j["value"] > 0
+
+
+def test_bench_torchrun_ws2_gloo():
+    """The driver's multi-rank launch path end to end (2 ranks, gloo on
+    CPU): rendezvous, per-rank pools, collectives inside the timed
+    region, MAX-over-ranks timing, one JSON line from rank 0 with the
+    whole-job aggregate."""
+    out = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+         "--master-port", "29877", os.path.join(REPO, "bench.py"),
+         "--gpus", "2", "--npsr", "2", "--ntoa", "120", "--ntm", "4",
+         "--rn-comps", "3", "--gwb-comps", "3", "--freqs", "8",
+         "--draws-per-step", "4", "--steps", "1", "--warmup", "0",
+         "--device", "cpu"],
+        cwd=REPO, capture_output=True, text=True, timeout=600,
+    )
+    assert out.returncode == 0, out.stderr[-2000:]
+    line = [l for l in out.stdout.strip().splitlines()
+            if l.startswith("{")][-1]
+    j = json.loads(line)
+    assert j["n_gpus"] == 2
+    assert j["config"]["global_batch"] == 8  # 2 ranks x 4 draws
+    assert j["config"]["spectrum_shape"] == [8, 8]  # gathered over ranks
