@@ -279,23 +279,6 @@ class NMFp:
                 fp[bad_idx] = self._direct_rows(engine, phiinvs, bad_idx,
                                                 draw_chunk, Nvecs, Ts)
                 return fp.cpu().numpy()
-        if os.environ.get("FASTFP_CLI_TIMING") == "1":
-            import time
-
-            torch.cuda.synchronize() if engine.device.type == "cuda" else None
-            t0 = time.perf_counter()
-            fp = engine.sweep(phiinvs=phiinvs, draw_chunk=draw_chunk)
-            torch.cuda.synchronize() if engine.device.type == "cuda" else None
-            t1 = time.perf_counter()
-            out = fp.cpu().numpy()
-            t2 = time.perf_counter()
-            ncomp = sum(b.comp is not None for b in engine.blocks)
-            print(f"[nmfp.sweep] engine.sweep {t1-t0:.3f} s, "
-                  f"to-host {t2-t1:.3f} s; stack="
-                  f"{getattr(engine, '_comp_stack', None) is not None} "
-                  f"comp={ncomp}/{len(engine.blocks)} chunk={draw_chunk}",
-                  flush=True)
-            return out
         fp = engine.sweep(phiinvs=phiinvs, draw_chunk=draw_chunk)
         return fp.cpu().numpy()
 
