@@ -648,8 +648,54 @@ class FpEngine:
         pinv_var = torch.stack(pin)  # (P, D, mv)
         fp_pp = torch.empty((P, min(draw_chunk, D), F),
                             dtype=torch.float64, device=self.device)
-        for lo in range(0, D, draw_chunk):
-            hi = min(lo + draw_chunk, D)
+        chunks = [(lo, min(lo + draw_chunk, D))
+                  for lo in range(0, D, draw_chunk)]
+        if self._use_hip and len(chunks) > 1:
+            # SOFTWARE PIPELINE across draw chunks: the Cholesky is
+            # latency-bound (~82% parked waves), the solve is
+            # throughput-bound, so chunk k+1's factor runs on a side
+            # stream UNDER chunk k's solve and its parked cycles are
+            # filled with solve work.  Factor tensors are event-fenced
+            # and record_stream'd so the caching allocator cannot
+            # recycle them across streams.
+            from fastfp_amd.ops import _fastfp_hip as ext
+
+            mvp = ops.pad16(st["G"].shape[-1])
+            main = torch.cuda.current_stream(self.device)
+            side = self._side_stream
+            side.wait_stream(main)  # pinv_var/delta0 visible to side
+
+            def factor(lo, hi):
+                with torch.cuda.stream(side):
+                    phi_var = (
+                        1.0 / (pinv_var[:, lo:hi, :]
+                               - st["delta0"][:, None, :])
+                    ).contiguous()
+                    L, invd = ext.chol_batch(st["G"], phi_var, mvp)
+                    ev = torch.cuda.Event()
+                    ev.record(side)
+                return L, invd, ev
+
+            pending = factor(*chunks[0])
+            for i, (lo, hi) in enumerate(chunks):
+                L, invd, ev = pending
+                if i + 1 < len(chunks):
+                    pending = factor(*chunks[i + 1])
+                main.wait_event(ev)
+                if hi - lo == fp_pp.shape[1]:
+                    pp = fp_pp
+                else:
+                    pp = torch.empty((P, hi - lo, F), dtype=torch.float64,
+                                     device=self.device)
+                pp.zero_()
+                ext.trsm_fp_accum(L, invd, st["K"], st["M0"], st["N0"],
+                                  pp, -1.0)
+                fp[lo:hi] += pp.sum(dim=0)
+                L.record_stream(main)
+                invd.record_stream(main)
+            return fp[0] if not batched else fp
+
+        for lo, hi in chunks:
             phi_var = (
                 1.0 / (pinv_var[:, lo:hi, :] - st["delta0"][:, None, :])
             ).contiguous()
