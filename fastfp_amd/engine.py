@@ -727,8 +727,48 @@ class FpEngine:
         pinv_all = torch.stack(pin)  # (P, D, m)
         fp_pp = torch.empty((P, min(draw_chunk, D), F),
                             dtype=torch.float64, device=self.device)
-        for lo in range(0, D, draw_chunk):
-            hi = min(lo + draw_chunk, D)
+        chunks = [(lo, min(lo + draw_chunk, D))
+                  for lo in range(0, D, draw_chunk)]
+        m = st["TNT"].shape[-1]
+        if self._use_hip and len(chunks) > 1 and ops.pad16(m) <= ops.MAX_MP_CHOL:
+            # same factor/solve software pipeline as the compressed
+            # path (the direct chol is an even larger step fraction)
+            from fastfp_amd.ops import _fastfp_hip as ext
+
+            mp = ops.pad16(m)
+            main = torch.cuda.current_stream(self.device)
+            side = self._side_stream
+            side.wait_stream(main)
+
+            def factor(lo, hi):
+                with torch.cuda.stream(side):
+                    L, invd = ext.chol_batch(
+                        st["TNT"], pinv_all[:, lo:hi, :].contiguous(), mp
+                    )
+                    ev = torch.cuda.Event()
+                    ev.record(side)
+                return L, invd, ev
+
+            pending = factor(*chunks[0])
+            for i, (lo, hi) in enumerate(chunks):
+                L, invd, ev = pending
+                if i + 1 < len(chunks):
+                    pending = factor(*chunks[i + 1])
+                main.wait_event(ev)
+                if hi - lo == fp_pp.shape[1]:
+                    pp = fp_pp
+                else:
+                    pp = torch.empty((P, hi - lo, F), dtype=torch.float64,
+                                     device=self.device)
+                pp.zero_()
+                ext.trsm_fp_accum(L, invd, st["RHS"], st["sNs"], st["sNr"],
+                                  pp, 1.0)
+                fp[lo:hi] += pp.sum(dim=0)
+                L.record_stream(main)
+                invd.record_stream(main)
+            return fp[0] if not batched else fp
+
+        for lo, hi in chunks:
             if hi - lo == fp_pp.shape[1]:
                 pp = fp_pp
             else:
