@@ -33,8 +33,9 @@ def test_bench_json_contract():
     assert j["config"]["global_batch"] == 4
 
 def test_bench_presets_parse():
-    j = _run("--preset", "ecorr67")
-    assert j["value"] > 0
+    for p in ("ecorr67", "fp45", "ska200", "nmfp67"):
+        j = _run("--preset", p)
+        assert j["value"] > 0, p
 
 
 def test_bench_torchrun_ws2_gloo():
