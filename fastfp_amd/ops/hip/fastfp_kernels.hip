@@ -537,7 +537,12 @@ __global__ __launch_bounds__(512, 4) void chol_batch_kernel(
 #define FPT_FREQS 63
 #define NBMAX (FASTFP_MAXMP / 16)
 
-template <int NBT, int DPG>
+// SWAP: launch with grid (D, ftiles, P) instead of (ftiles, D, P) so
+// consecutive workgroups share one RHS column strip (L2-resident
+// across ~D consecutive blocks) instead of streaming 16 different
+// strips — A/B arm for the measured 2.9 ms RHS-load attribution
+// (FASTFP_TRSM_GRID=d).
+template <int NBT, int DPG, bool SWAP = false>
 __global__ __launch_bounds__(512, 4) void trsm_fp_kernel(
     const double* __restrict__ L_all /*(P*D,mp,mp)*/,
     const double* __restrict__ invd_all /*(P*D, mp/16, 16, 16)*/,
@@ -569,8 +574,8 @@ __global__ __launch_bounds__(512, 4) void trsm_fp_kernel(
   const double* sNs = sNs_all + (long)pp * 3 * F;
   const double* sNr = sNr_all + (long)pp * 2 * F;
   double* fp = fp_all + (long)pp * D * F;
-  const int d0 = blockIdx.y * DPG;
-  const int f0 = blockIdx.x * FPT_FREQS;
+  const int d0 = (SWAP ? blockIdx.x : blockIdx.y) * DPG;
+  const int f0 = (SWAP ? blockIdx.y : blockIdx.x) * FPT_FREQS;
   const int tid = threadIdx.x;
   const int lane = tid & 63;
   const int wv = tid >> 6;
@@ -1298,6 +1303,22 @@ void launch_trsm_fp(const double* L, const double* invd, const double* RHS,
       TRSM_RL_CASE(1) TRSM_RL_CASE(2) TRSM_RL_CASE(3) TRSM_RL_CASE(4)
       TRSM_RL_CASE(5) TRSM_RL_CASE(6) TRSM_RL_CASE(7) TRSM_RL_CASE(8)
 #undef TRSM_RL_CASE
+    }
+    return;
+  }
+  // FASTFP_TRSM_GRID=d: draws-innermost grid (RHS L2-locality A/B arm)
+  static const char* grid_env = getenv("FASTFP_TRSM_GRID");
+  static const bool swap_grid = grid_env && grid_env[0] == 'd';
+  if (swap_grid && nb <= 8) {
+    const dim3 grid(D, ftiles, P);
+    switch (nb) {
+#define TRSM_SW_CASE(NBT) \
+      case NBT: hipLaunchKernelGGL((trsm_fp_kernel<NBT, 1, true>), grid, \
+                    blk, 0, stream, L, invd, RHS, sNs, sNr, F, D, gsign, \
+                    fp); break;
+      TRSM_SW_CASE(1) TRSM_SW_CASE(2) TRSM_SW_CASE(3) TRSM_SW_CASE(4)
+      TRSM_SW_CASE(5) TRSM_SW_CASE(6) TRSM_SW_CASE(7) TRSM_SW_CASE(8)
+#undef TRSM_SW_CASE
     }
     return;
   }
