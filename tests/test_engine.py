@@ -1,6 +1,7 @@
 """Engine correctness: the restructured sweep vs the dense oracle and vs
 the reference-shaped parity path."""
 
+import torch
 import numpy as np
 import pytest
 
@@ -313,6 +314,74 @@ def test_per_draw_hybrid_split_matches_direct():
                       compress=False)
     assert np.isfinite(got).all()
     np.testing.assert_allclose(got, direct, rtol=1e-7)
+
+
+def test_force_direct_bypasses_compression():
+    """engine.sweep(force_direct=True) must ignore an enabled
+    compression and reproduce the plain direct sweep exactly."""
+    psrs = make_synthetic_pta(npsr=2, ntoa=300, ntm=6, seed=18)
+    noise = {"gw_gamma": 13.0 / 3.0, "gw_log10_A": float(np.log10(2e-15))}
+    for p in psrs:
+        noise[f"{p.name}_red_noise_gamma"] = 4.0
+        noise[f"{p.name}_red_noise_log10_A"] = -14.5
+    pta = initialize_pta(psrs, noise, inc_cp=True, rn_comps=8, gwb_comps=8)
+    TNTs, Nvecs, Ts = get_mats_nmfp(pta, noise)
+    D = 3
+    rng = np.random.default_rng(4)
+    phiinvs = [
+        c.get_phiinv({
+            n: (rng.uniform(2, 6, D) if n.endswith("gamma")
+                else rng.uniform(-16, -14, D))
+            for n in pta.params
+        })
+        for c in pta.rn_containers
+    ]
+    freqs = np.linspace(3e-9, 5e-8, 7)
+    eng_plain = FpEngine(psrs, Nvecs, Ts, device="cpu").precompute(freqs)
+    want = eng_plain.sweep(phiinvs=phiinvs).numpy()
+    eng = FpEngine(psrs, Nvecs, Ts, device="cpu").precompute(freqs)
+    eng.enable_draw_compression(
+        [c.var_slice for c in pta.rn_containers],
+        [c.get_phiinv(noise) for c in pta.rn_containers],
+    )
+    assert all(b.comp is not None for b in eng.blocks)
+    got = eng.sweep(phiinvs=phiinvs, force_direct=True).numpy()
+    np.testing.assert_array_equal(got, want)
+
+
+def test_per_draw_margin_vector():
+    """compression_margin_per_draw: per-draw mins, +inf when nothing
+    compressed, consistent with the scalar margin."""
+    psrs = make_synthetic_pta(npsr=2, ntoa=200, ntm=5, seed=19)
+    noise = {"gw_gamma": 13.0 / 3.0, "gw_log10_A": float(np.log10(2e-15))}
+    for p in psrs:
+        noise[f"{p.name}_red_noise_gamma"] = 4.0
+        noise[f"{p.name}_red_noise_log10_A"] = -14.5
+    pta = initialize_pta(psrs, noise, inc_cp=True, rn_comps=5, gwb_comps=5)
+    TNTs, Nvecs, Ts = get_mats_nmfp(pta, noise)
+    freqs = np.linspace(3e-9, 5e-8, 4)
+    eng = FpEngine(psrs, Nvecs, Ts, device="cpu").precompute(freqs)
+    D = 4
+    rng = np.random.default_rng(6)
+    samples = {
+        n: (rng.uniform(3, 5, D) if n.endswith("gamma")
+            else rng.uniform(-15.5, -14.5, D))
+        for n in pta.params
+    }
+    from fastfp_amd.noise import batch_phiinv
+
+    piv = [p if p.dim() == 2 else p[None]
+           for p in batch_phiinv(pta.rn_containers, samples)]
+    # nothing compressed yet -> +inf margins
+    m0 = eng.compression_margin_per_draw(piv)
+    assert m0.shape == (D,) and bool(torch.isinf(m0).all())
+    eng.enable_draw_compression(
+        [c.var_slice for c in pta.rn_containers],
+        [c.get_phiinv(noise) for c in pta.rn_containers],
+    )
+    md = eng.compression_margin_per_draw(piv)
+    assert md.shape == (D,) and bool((md > 0).all())
+    assert float(md.min()) == pytest.approx(eng.compression_margin(piv))
 
 
 def test_compression_margin_fallback():
