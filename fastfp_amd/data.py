@@ -98,6 +98,50 @@ class PulsarData:
             backend_flags=z["backend_flags"].astype(object),
         )
 
+    def save_feather(self, path: str) -> None:
+        """Write the pulsar as an Arrow/feather table (one row per TOA:
+        toas/toaerrs/residuals/backend_flags columns plus the design
+        matrix as ``Mmat_j`` columns; pulsar name in the schema
+        metadata).  This is fastfp_amd's own self-describing feather
+        schema, not enterprise's internal FeatherPulsar layout (whose
+        spec is not public); ``load_pulsars`` round-trips it exactly."""
+        import pyarrow as pa
+        import pyarrow.feather as paf
+
+        cols = {
+            "toas": self.toas,
+            "toaerrs": self.toaerrs,
+            "residuals": self.residuals,
+            "backend_flags": np.asarray(self.backend_flags, dtype=str),
+        }
+        for j in range(self.ntm):
+            cols[f"Mmat_{j}"] = self.Mmat[:, j]
+        table = pa.table(cols).replace_schema_metadata(
+            {"fastfp_amd.name": self.name, "fastfp_amd.ntm": str(self.ntm)}
+        )
+        paf.write_feather(table, path)
+
+    @classmethod
+    def load_feather(cls, path: str) -> "PulsarData":
+        import pyarrow.feather as paf
+
+        table = paf.read_table(path)
+        meta = table.schema.metadata or {}
+        name = meta.get(b"fastfp_amd.name", b"unknown").decode()
+        ntm = int(meta.get(b"fastfp_amd.ntm", b"0"))
+        cols = {c: table[c].to_numpy() for c in table.column_names}
+        Mmat = np.stack(
+            [cols[f"Mmat_{j}"] for j in range(ntm)], axis=1
+        ) if ntm else np.zeros((len(cols["toas"]), 0))
+        return cls(
+            name=name,
+            toas=cols["toas"],
+            toaerrs=cols["toaerrs"],
+            residuals=cols["residuals"],
+            Mmat=Mmat,
+            backend_flags=cols["backend_flags"].astype(object),
+        )
+
     @classmethod
     def from_object(cls, obj) -> "PulsarData":
         """Duck-typed converter from an enterprise-style Pulsar object
@@ -125,11 +169,21 @@ def load_pulsars(path: str) -> list:
     - a ``.pkl`` pickle of a list of enterprise-style Pulsar objects
       (the reference's input format, ``/root/reference/examples/run_fp.py:34``),
     - a ``.npz`` file saved by :func:`save_pulsars`,
-    - a directory of per-pulsar ``.npz`` files.
+    - a per-pulsar ``.feather`` file (:meth:`PulsarData.save_feather`),
+    - a directory of per-pulsar ``.npz`` and/or ``.feather`` files.
     """
     if os.path.isdir(path):
-        files = sorted(glob.glob(os.path.join(path, "*.npz")))
-        return [PulsarData.load_npz(f) for f in files]
+        out = [
+            PulsarData.load_npz(f)
+            for f in sorted(glob.glob(os.path.join(path, "*.npz")))
+        ]
+        out += [
+            PulsarData.load_feather(f)
+            for f in sorted(glob.glob(os.path.join(path, "*.feather")))
+        ]
+        return out
+    if path.endswith(".feather"):
+        return [PulsarData.load_feather(path)]
     if path.endswith(".pkl") or path.endswith(".pickle"):
         with open(path, "rb") as f:
             objs = pickle.load(f)

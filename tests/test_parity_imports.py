@@ -45,3 +45,26 @@ def test_enterprise_style_pickle_loads(tmp_path):
     psrs = load_pulsars(path)
     assert len(psrs) == 1 and psrs[0].ntoa == 40
     assert psrs[0].name == "J0000+0000"
+
+
+def test_feather_roundtrip(tmp_path):
+    """Per-pulsar feather save/load round-trips exactly (the SURVEY §7
+    npz/feather loader plan; fastfp_amd's own self-describing schema)."""
+    from fastfp_amd.data import PulsarData, load_pulsars, make_synthetic_pta
+
+    psrs = make_synthetic_pta(npsr=2, ntoa=50, ntm=4, seed=3)
+    for p in psrs:
+        p.save_feather(str(tmp_path / f"{p.name}.feather"))
+    back = load_pulsars(str(tmp_path))
+    assert len(back) == 2
+    by_name = {p.name: p for p in back}
+    for p in psrs:
+        q = by_name[p.name]
+        np.testing.assert_array_equal(q.toas, p.toas)
+        np.testing.assert_array_equal(q.toaerrs, p.toaerrs)
+        np.testing.assert_array_equal(q.residuals, p.residuals)
+        np.testing.assert_array_equal(q.Mmat, p.Mmat)
+        assert list(q.backend_flags) == list(p.backend_flags)
+    # single-file load too
+    one = load_pulsars(str(tmp_path / f"{psrs[0].name}.feather"))
+    assert len(one) == 1 and one[0].ntoa == psrs[0].ntoa
