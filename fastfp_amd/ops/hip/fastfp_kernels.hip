@@ -510,6 +510,66 @@ __global__ __launch_bounds__(512, 4) void chol_batch_kernel(
 }
 
 // ---------------------------------------------------------------------
+// chol_wave: fully IN-WAVE Cholesky for mp <= 64 — one DRAW per wave,
+// rows as lanes, the whole factor in registers with width-64 __shfl
+// broadcasts.  No LDS, no barriers, no cross-wave dependencies: every
+// wave is busy on its own draw the whole time, attacking the blocked
+// kernel's 82%-parked-waves structure (waves idling on the serial
+// 16x16 diagonal) at its root.  The 16x16 diagonal-block inverses the
+// solver needs are produced by the separate diag_inv kernel (reused
+// from the m > 128 path).  The t/j loops are fully unrolled so every
+// a[] index is compile-time (runtime indices demote the register array
+// to scratch).  grid = (ceil(D/4), P), block = 256 (4 waves).
+// Selected by FASTFP_CHOL_ALGO=wave (A/B arm).
+// ---------------------------------------------------------------------
+template <int MP>
+__global__ __launch_bounds__(256) void chol_wave_kernel(
+    const double* __restrict__ TNT_all /*(P,m,m)*/,
+    const double* __restrict__ phiinv_all /*(P,D,m)*/, int m, int D,
+    double* __restrict__ L_all /*(P*D,mp,mp)*/) {
+  const int pp = blockIdx.y;
+  const int d = blockIdx.x * 4 + (threadIdx.x >> 6);
+  if (d >= D) return;
+  const int i = threadIdx.x & 63;  // row owned by this lane
+  const double* TNT = TNT_all + (long)pp * m * m;
+  const double* phiinv = phiinv_all + ((long)pp * D + d) * m;
+  double* L = L_all + ((long)pp * D + d) * MP * MP;
+
+  // row i of Sigma (lower triangle only; padding rows are identity)
+  double a[MP];
+#pragma unroll
+  for (int j = 0; j < MP; ++j) {
+    double v = 0.0;
+    if (i < m && j < m && j <= i) v = TNT[(long)i * m + j];
+    if (j == i) v += (i < m) ? phiinv[i] : 1.0;
+    a[j] = (i < MP) ? v : 0.0;
+  }
+
+  // unblocked right-looking factor, rows-as-lanes
+#pragma unroll
+  for (int t = 0; t < MP; ++t) {
+    const double att = __shfl(a[t], t, 64);
+    const double rdv = 1.0 / sqrt(att);
+    if (i == t)
+      a[t] = att * rdv;  // = sqrt(att)
+    else if (i > t)
+      a[t] *= rdv;
+#pragma unroll
+    for (int j = t + 1; j < MP; ++j) {
+      const double ljt = __shfl(a[t], j, 64);
+      if (i >= j) a[j] = fma(-a[t], ljt, a[j]);
+    }
+  }
+
+  // write row i (upper half as deterministic zeros)
+  if (i < MP) {
+#pragma unroll
+    for (int j = 0; j < MP; ++j)
+      L[(long)i * MP + j] = (j <= i) ? a[j] : 0.0;
+  }
+}
+
+// ---------------------------------------------------------------------
 // trsm_fp: per (draw, frequency-tile) forward-substitute
 //   W = L^-1 [B_cols | u]   (u = the T^T N^-1 r column, solved
 //   redundantly per tile: one column vs 126)
@@ -1222,12 +1282,32 @@ void launch_sbgemm(const double* T, const double* toas, const double* ninv,
                      plane_stride, ldo);
 }
 
+void launch_diag_inv(const double* L, int mp, long nblk_total, double* invd,
+                     hipStream_t stream);
+
 void launch_chol_batch(const double* TNT, const double* phiinv, int m, int mp,
                        int D, int P, double* L, double* invd,
                        hipStream_t stream) {
   // small matrices: 256 threads -> 4 workgroups/CU despite the 115-VGPR
   // diagonal-factor pressure; large: 512 threads for MFMA coverage
   const dim3 grid(D, P), blk(mp <= 64 ? 256 : 512);
+  // FASTFP_CHOL_ALGO=wave: fully in-wave factor (one draw per wave,
+  // no LDS/barriers) for mp <= 64; the diagonal-block inverses come
+  // from the diag_inv kernel
+  static const char* walgo = getenv("FASTFP_CHOL_ALGO");
+  static const bool use_wave = walgo && 0 == __builtin_strcmp(walgo, "wave");
+  if (use_wave && mp <= 64) {
+    const dim3 wgrid((D + 3) / 4, P);
+    switch (mp >> 4) {
+#define CHOL_WAVE_CASE(NBT) \
+      case NBT: hipLaunchKernelGGL((chol_wave_kernel<NBT * 16>), wgrid, \
+                    dim3(256), 0, stream, TNT, phiinv, m, D, L); break;
+      CHOL_WAVE_CASE(1) CHOL_WAVE_CASE(2) CHOL_WAVE_CASE(3) CHOL_WAVE_CASE(4)
+#undef CHOL_WAVE_CASE
+    }
+    launch_diag_inv(L, mp, (long)P * D * (mp >> 4), invd, stream);
+    return;
+  }
   // Lower-triangle-packed Sigma is the DEFAULT (measured -13% on chol
   // at the bench shape from the extra workgroup of concurrency; at
   // mp=128 the square layout is 132 KB LDS = ONE workgroup per CU, so
