@@ -513,14 +513,19 @@ __global__ __launch_bounds__(512, 4) void chol_batch_kernel(
 // chol_wave: fully IN-WAVE Cholesky for mp <= 64 — one DRAW per wave,
 // rows as lanes, the whole factor in registers with width-64 __shfl
 // broadcasts.  No LDS, no barriers, no cross-wave dependencies: every
-// wave is busy on its own draw the whole time, attacking the blocked
-// kernel's 82%-parked-waves structure (waves idling on the serial
-// 16x16 diagonal) at its root.  The 16x16 diagonal-block inverses the
-// solver needs are produced by the separate diag_inv kernel (reused
-// from the m > 128 path).  The t/j loops are fully unrolled so every
-// a[] index is compile-time (runtime indices demote the register array
-// to scratch).  grid = (ceil(D/4), P), block = 256 (4 waves).
-// Selected by FASTFP_CHOL_ALGO=wave (A/B arm).
+// wave is busy on its own draw, attacking the blocked kernel's
+// 82%-parked-waves structure at its root.  The diagonal-block
+// inverses come from the separate diag_inv kernel.  The t/j loops are
+// fully unrolled so every a[] index is compile-time (runtime indices
+// demote the register array to scratch).
+//
+// MEASURED NEGATIVE (kept as the record): 4.90 ms vs the blocked
+// kernel's 3.04 ms at the bench shape.  The ~2,080 serial-ish __shfl
+// broadcasts per draw (DS-pipe) plus 209 VGPR -> 2 waves/SIMD swamp
+// the parking win; the blocked kernel's MFMA trailing updates do the
+// same O(mp^3) work in far fewer issue slots.  Numerics validated
+// (7 GPU tests green under FASTFP_CHOL_ALGO=wave).
+// grid = (ceil(D/4), P), block = 256 (4 waves).  A/B arm only.
 // ---------------------------------------------------------------------
 template <int MP>
 __global__ __launch_bounds__(256) void chol_wave_kernel(
