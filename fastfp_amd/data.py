@@ -106,7 +106,7 @@ class PulsarData:
         schema, not enterprise's internal FeatherPulsar layout (whose
         spec is not public); ``load_pulsars`` round-trips it exactly."""
         import pyarrow as pa
-        import pyarrow.feather as paf
+        import pyarrow.ipc as paipc
 
         cols = {
             "toas": self.toas,
@@ -119,13 +119,16 @@ class PulsarData:
         table = pa.table(cols).replace_schema_metadata(
             {"fastfp_amd.name": self.name, "fastfp_amd.ntm": str(self.ntm)}
         )
-        paf.write_feather(table, path)
+        # the Arrow IPC file format IS Feather V2
+        with paipc.new_file(path, table.schema) as w:
+            w.write_table(table)
 
     @classmethod
     def load_feather(cls, path: str) -> "PulsarData":
-        import pyarrow.feather as paf
+        import pyarrow.ipc as paipc
 
-        table = paf.read_table(path)
+        with paipc.open_file(path) as r:
+            table = r.read_all()
         meta = table.schema.metadata or {}
         name = meta.get(b"fastfp_amd.name", b"unknown").decode()
         ntm = int(meta.get(b"fastfp_amd.ntm", b"0"))
