@@ -632,6 +632,69 @@ class FpEngine:
 
         return fp[0] if not batched else fp
 
+    def _pipeline_chunks(self, chunks, P, mp, F, fp, fp_pp, factor_into,
+                         solve):
+        """Factor/solve software pipeline over draw chunks with
+        PING-PONG factor buffers and two-way event fencing.
+
+        Chunk k+1's factor (side stream, buffer (k+1)&1) overlaps chunk
+        k's solve (main stream); before overwriting a buffer, the side
+        stream waits on the event recorded after the solve that last
+        read it.  Explicit reusable buffers instead of per-chunk
+        allocations + record_stream: the latter defers caching-
+        allocator reuse across streams and measured as multi-GB
+        allocator churn (2.2 s/step) on the 9.5 GB direct-path chunks.
+        The structure is hipGraph-capturable (events + two streams
+        forked from the capture stream)."""
+        dev = self.device
+        chunk0 = chunks[0][1] - chunks[0][0]
+        nmax = P * chunk0
+        key = ("pipebuf", nmax, mp)
+        bufs = getattr(self, "_pipe_bufs", None)
+        if bufs is None or bufs[0] != key:
+            L0 = torch.empty((nmax, mp, mp), dtype=torch.float64, device=dev)
+            L1 = torch.empty_like(L0)
+            i0 = torch.empty((nmax, mp // 16, 16, 16), dtype=torch.float64,
+                             device=dev)
+            i1 = torch.empty_like(i0)
+            bufs = (key, (L0, L1), (i0, i1))
+            self._pipe_bufs = bufs
+        _, Lb, Ib = bufs
+
+        main = torch.cuda.current_stream(dev)
+        side = self._side_stream
+        side.wait_stream(main)  # inputs visible to side
+        solve_done = [None, None]  # event after the solve reading buf b
+
+        def factor(i, lo, hi):
+            b = i & 1
+            with torch.cuda.stream(side):
+                if solve_done[b] is not None:
+                    side.wait_event(solve_done[b])
+                factor_into(Lb[b], Ib[b], lo, hi)
+                ev = torch.cuda.Event()
+                ev.record(side)
+            return ev
+
+        pending = factor(0, *chunks[0])
+        for i, (lo, hi) in enumerate(chunks):
+            ev = pending
+            if i + 1 < len(chunks):
+                pending = factor(i + 1, *chunks[i + 1])
+            main.wait_event(ev)
+            b = i & 1
+            if hi - lo == fp_pp.shape[1]:
+                pp = fp_pp
+            else:
+                pp = torch.empty((P, hi - lo, F), dtype=torch.float64,
+                                 device=dev)
+            pp.zero_()
+            solve(Lb[b], Ib[b], lo, hi, pp)
+            fp[lo:hi] += pp.sum(dim=0)
+            ev2 = torch.cuda.Event()
+            ev2.record(main)
+            solve_done[b] = ev2
+
     def _sweep_stacked(self, phiinvs, fp, D, F, draw_chunk, batched):
         """Pulsar-batched compressed sweep: one chol + one trsm launch
         per draw chunk across ALL pulsars."""
@@ -655,44 +718,29 @@ class FpEngine:
             # latency-bound (~82% parked waves), the solve is
             # throughput-bound, so chunk k+1's factor runs on a side
             # stream UNDER chunk k's solve and its parked cycles are
-            # filled with solve work.  Factor tensors are event-fenced
-            # and record_stream'd so the caching allocator cannot
-            # recycle them across streams.
-            from fastfp_amd.ops import _fastfp_hip as ext
-
+            # filled with solve work.
             mvp = ops.pad16(st["G"].shape[-1])
-            main = torch.cuda.current_stream(self.device)
-            side = self._side_stream
-            side.wait_stream(main)  # pinv_var/delta0 visible to side
 
-            def factor(lo, hi):
-                with torch.cuda.stream(side):
-                    phi_var = (
-                        1.0 / (pinv_var[:, lo:hi, :]
-                               - st["delta0"][:, None, :])
-                    ).contiguous()
-                    L, invd = ext.chol_batch(st["G"], phi_var, mvp)
-                    ev = torch.cuda.Event()
-                    ev.record(side)
-                return L, invd, ev
+            def factor_into(L, invd, lo, hi):
+                from fastfp_amd.ops import _fastfp_hip as ext
 
-            pending = factor(*chunks[0])
-            for i, (lo, hi) in enumerate(chunks):
-                L, invd, ev = pending
-                if i + 1 < len(chunks):
-                    pending = factor(*chunks[i + 1])
-                main.wait_event(ev)
-                if hi - lo == fp_pp.shape[1]:
-                    pp = fp_pp
-                else:
-                    pp = torch.empty((P, hi - lo, F), dtype=torch.float64,
-                                     device=self.device)
-                pp.zero_()
-                ext.trsm_fp_accum(L, invd, st["K"], st["M0"], st["N0"],
-                                  pp, -1.0)
-                fp[lo:hi] += pp.sum(dim=0)
-                L.record_stream(main)
-                invd.record_stream(main)
+                phi_var = (
+                    1.0 / (pinv_var[:, lo:hi, :]
+                           - st["delta0"][:, None, :])
+                ).contiguous()
+                n = P * (hi - lo)
+                ext.chol_batch_into(st["G"], phi_var, mvp,
+                                    L[:n], invd[:n])
+
+            def solve(L, invd, lo, hi, pp):
+                from fastfp_amd.ops import _fastfp_hip as ext
+
+                n = P * (hi - lo)
+                ext.trsm_fp_accum(L[:n], invd[:n], st["K"], st["M0"],
+                                  st["N0"], pp, -1.0)
+
+            self._pipeline_chunks(chunks, P, mvp, F, fp, fp_pp,
+                                  factor_into, solve)
             return fp[0] if not batched else fp
 
         for lo, hi in chunks:
@@ -731,41 +779,27 @@ class FpEngine:
                   for lo in range(0, D, draw_chunk)]
         m = st["TNT"].shape[-1]
         if self._use_hip and len(chunks) > 1 and ops.pad16(m) <= ops.MAX_MP_CHOL:
-            # same factor/solve software pipeline as the compressed
-            # path (the direct chol is an even larger step fraction)
-            from fastfp_amd.ops import _fastfp_hip as ext
-
+            # same factor/solve software pipeline as the compressed path
             mp = ops.pad16(m)
-            main = torch.cuda.current_stream(self.device)
-            side = self._side_stream
-            side.wait_stream(main)
 
-            def factor(lo, hi):
-                with torch.cuda.stream(side):
-                    L, invd = ext.chol_batch(
-                        st["TNT"], pinv_all[:, lo:hi, :].contiguous(), mp
-                    )
-                    ev = torch.cuda.Event()
-                    ev.record(side)
-                return L, invd, ev
+            def factor_into(L, invd, lo, hi):
+                from fastfp_amd.ops import _fastfp_hip as ext
 
-            pending = factor(*chunks[0])
-            for i, (lo, hi) in enumerate(chunks):
-                L, invd, ev = pending
-                if i + 1 < len(chunks):
-                    pending = factor(*chunks[i + 1])
-                main.wait_event(ev)
-                if hi - lo == fp_pp.shape[1]:
-                    pp = fp_pp
-                else:
-                    pp = torch.empty((P, hi - lo, F), dtype=torch.float64,
-                                     device=self.device)
-                pp.zero_()
-                ext.trsm_fp_accum(L, invd, st["RHS"], st["sNs"], st["sNr"],
-                                  pp, 1.0)
-                fp[lo:hi] += pp.sum(dim=0)
-                L.record_stream(main)
-                invd.record_stream(main)
+                n = P * (hi - lo)
+                ext.chol_batch_into(
+                    st["TNT"], pinv_all[:, lo:hi, :].contiguous(), mp,
+                    L[:n], invd[:n],
+                )
+
+            def solve(L, invd, lo, hi, pp):
+                from fastfp_amd.ops import _fastfp_hip as ext
+
+                n = P * (hi - lo)
+                ext.trsm_fp_accum(L[:n], invd[:n], st["RHS"], st["sNs"],
+                                  st["sNr"], pp, 1.0)
+
+            self._pipeline_chunks(chunks, P, mp, F, fp, fp_pp,
+                                  factor_into, solve)
             return fp[0] if not batched else fp
 
         for lo, hi in chunks:

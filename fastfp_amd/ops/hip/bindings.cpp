@@ -99,6 +99,31 @@ std::vector<torch::Tensor> chol_batch(torch::Tensor TNT, torch::Tensor phiinv,
   return {L, invd};
 }
 
+// chol_batch_into: same as chol_batch but writing into caller-owned
+// L/invd (the engine's ping-pong pipeline buffers — avoids per-chunk
+// allocator churn across streams).
+void chol_batch_into(torch::Tensor TNT, torch::Tensor phiinv, int64_t mp,
+                     torch::Tensor L, torch::Tensor invd) {
+  check_f64(TNT, "TNT");
+  check_f64(phiinv, "phiinv");
+  check_f64(L, "L");
+  check_f64(invd, "invd");
+  const bool batched = TNT.dim() == 3;
+  const int P = batched ? TNT.size(0) : 1;
+  const int m = TNT.size(batched ? 1 : 0);
+  const int D = phiinv.size(batched ? 1 : 0);
+  TORCH_CHECK(phiinv.dim() == (batched ? 3 : 2), "phiinv rank");
+  TORCH_CHECK(phiinv.size(batched ? 2 : 1) == m, "phiinv width != m");
+  TORCH_CHECK(mp % 16 == 0 && mp <= 128 && mp >= m, "mp");
+  TORCH_CHECK(L.size(0) == (long)P * D && L.size(1) == mp && L.size(2) == mp,
+              "L shape");
+  TORCH_CHECK(invd.size(0) == (long)P * D && invd.size(1) == mp / 16,
+              "invd shape");
+  launch_chol_batch(TNT.data_ptr<double>(), phiinv.data_ptr<double>(), m,
+                    (int)mp, D, P, L.data_ptr<double>(),
+                    invd.data_ptr<double>(), stream());
+}
+
 // trsm_fp_accum: single pulsar (L (D,mp,mp), RHS (mp,2F+1), sNs (3,F),
 // fp (D,F)) or pulsar-batched (L (P*D,mp,mp), RHS (P,mp,2F+1),
 // sNs (P,3,F), fp (P,D,F)); accumulates in place.
@@ -178,6 +203,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("sigdots", &sigdots, "fused sincos signal-basis dots");
   m.def("sbgemm", &sbgemm, "fused signal-basis MFMA fp64 DGEMM");
   m.def("chol_batch", &chol_batch, "batched LDS-resident fp64 Cholesky");
+  m.def("chol_batch_into", &chol_batch_into,
+        "chol_batch into caller-owned L/invd (pipeline buffers)");
   m.def("trsm_fp_accum", &trsm_fp_accum,
         "batched TRSM + fused 2x2 Fp reduction");
 }
