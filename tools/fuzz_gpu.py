@@ -83,8 +83,12 @@ def one_case(seed):
     # is conditioning-amplified to ~1e-3 relative at cancellation-heavy
     # frequencies, so the injected tolerance checks STRUCTURE (a
     # misrouted hybrid row would be O(1) wrong), not round-off
+    # (atol floor for injected cases: cancellation-dominated tiny Fp
+    # values carry O(1e-4) absolute round-off on the near-singular row
+    # — docs/DESIGN.md §8; a mis-routed row would be O(1) wrong)
     np.testing.assert_allclose(gpu, cpu, rtol=1e-2 if injected else 1e-4,
-                               atol=1e-9, err_msg=desc)
+                               atol=1e-3 if injected else 1e-9,
+                               err_msg=desc)
 
     # plain Fp path too
     Nvecs2, Ts2, sigmas = get_mats_fp(pta, noise)
