@@ -60,3 +60,27 @@ def test_bench_torchrun_ws2_gloo():
     assert j["n_gpus"] == 2
     assert j["config"]["global_batch"] == 8  # 2 ranks x 4 draws
     assert j["config"]["spectrum_shape"] == [8, 8]  # gathered over ranks
+
+
+def test_bench_torchrun_ws4_gloo():
+    """4 ranks — the driver's N=4 scaling launch shape.  Exercises the
+    >2-rank collective path (size exchange + padded all-gather over 4
+    shards, MAX-over-ranks timing) that the 4- and 8-GPU round-end runs
+    hit."""
+    out = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "4", "--master-addr", "127.0.0.1",
+         "--master-port", "29878", os.path.join(REPO, "bench.py"),
+         "--gpus", "4", "--npsr", "2", "--ntoa", "120", "--ntm", "4",
+         "--rn-comps", "3", "--gwb-comps", "3", "--freqs", "8",
+         "--draws-per-step", "3", "--steps", "1", "--warmup", "0",
+         "--device", "cpu"],
+        cwd=REPO, capture_output=True, text=True, timeout=600,
+    )
+    assert out.returncode == 0, out.stderr[-2000:]
+    line = [l for l in out.stdout.strip().splitlines()
+            if l.startswith("{")][-1]
+    j = json.loads(line)
+    assert j["n_gpus"] == 4
+    assert j["config"]["global_batch"] == 12  # 4 ranks x 3 draws
+    assert j["config"]["spectrum_shape"] == [12, 8]  # gathered over ranks
