@@ -81,7 +81,7 @@ class NMFp:
         self.toas = [np.asarray(p.toas, dtype=np.float64) for p in psrs]
         self.residuals = [np.asarray(p.residuals, dtype=np.float64) for p in psrs]
         self._phi_homog = None  # cached check_batch_homogeneous result
-        self._graphs = {}  # (D, draw_chunk, names) -> _GraphedEngineSweep
+        self._graphs = {}  # (id(engine), D, draw_chunk, names) -> _GraphedEngineSweep
         self._graph_seen = {}  # shape-repeat counts (capture on 2nd)
 
     def __call__(self, fgw, samples, Nvecs, Ts, TNTs):
@@ -174,12 +174,23 @@ class NMFp:
         D = len(vals[0])
         if any(len(v) != D for v in vals):
             return None
-        key = (D, draw_chunk, tuple(names))
+        # the captured graph closes over ONE engine (its freq grid, its
+        # compression stack), so the engine's identity must be part of
+        # the key: a same-shaped sweep against a different engine (new
+        # freqs, same D) must not replay a stale capture.  Cached
+        # entries hold a strong reference to their engine, so a live
+        # id() here can only name that same object.  engine=None
+        # callers get a fresh engine per sweep and therefore never
+        # reuse a capture — they stay on the eager path (correct: their
+        # per-call precompute dwarfs the launch overhead anyway).
+        key = (id(engine), D, draw_chunk, tuple(names))
         g = self._graphs.get(key)
         if g is None:
             # capture costs ~2 sweeps + graph instantiation, so only
             # capture once a shape REPEATS (batch 2 of a CLI loop);
             # one-shot sweeps stay eager
+            if len(self._graph_seen) > 64:  # engine-per-call callers
+                self._graph_seen.clear()    # never repeat a key: bound it
             seen = self._graph_seen.get(key, 0) + 1
             self._graph_seen[key] = seen
             if seen < 2 or len(self._graphs) >= 4:  # bound graph pool

@@ -576,3 +576,39 @@ def test_precompute_invalidates_stale_compression():
     eng.precompute(np.linspace(4e-9, 5e-8, 9))
     assert all(blk.comp is None for blk in eng.blocks)
     assert eng._comp_stack is None
+
+
+def test_graph_cache_keyed_by_engine_identity():
+    """The hipGraph CLI cache closes over one engine (its freq grid),
+    so a same-shaped sweep against a DIFFERENT engine must not replay
+    a stale capture: the cache key includes the engine's identity.
+    Exercised with stub engines that satisfy the graph-path
+    preconditions (the capture itself is HIP-only and fails closed)."""
+    from fastfp_amd.nmfp import NMFp
+
+    class StubEngine:
+        _use_hip = True
+        _comp_stack = object()
+        device = "cpu"
+
+    nm = NMFp.__new__(NMFp)
+    nm.rn_sigs = []
+    nm._phi_homog = True
+    nm._graphs = {}
+    nm._graph_seen = {}
+    samples = {"gw_gamma": np.ones(3), "gw_log10_A": np.full(3, -14.5)}
+
+    eng_a, eng_b = StubEngine(), StubEngine()
+    # first sighting of (engine A, shape): eager
+    assert nm._sweep_graphed(eng_a, samples, 4, True, None, None) is None
+    key_a = (id(eng_a), 3, 4, tuple(samples))
+    assert nm._graph_seen[key_a] == 1 and key_a not in nm._graphs
+    # second sighting: capture attempted; stub engine -> permanent
+    # eager fallback CACHED UNDER ENGINE A's KEY
+    assert nm._sweep_graphed(eng_a, samples, 4, True, None, None) is None
+    assert nm._graphs[key_a] is False
+    # engine B, same shapes: must NOT hit engine A's cache entry
+    assert nm._sweep_graphed(eng_b, samples, 4, True, None, None) is None
+    key_b = (id(eng_b), 3, 4, tuple(samples))
+    assert key_b != key_a
+    assert nm._graph_seen[key_b] == 1 and key_b not in nm._graphs
