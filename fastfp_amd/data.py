@@ -153,13 +153,17 @@ class PulsarData:
         toaerrs = getattr(obj, "toaerrs", None)
         if toaerrs is None:
             toaerrs = np.full(len(obj.toas), 1e-6)
+        # keep a MISSING backend_flags as None (np.asarray(None) is a
+        # 0-d object array, which would bypass the per-TOA default in
+        # __post_init__ and break backend masking downstream)
+        bflags = getattr(obj, "backend_flags", None)
         return cls(
             name=str(obj.name),
             toas=np.asarray(obj.toas, dtype=np.float64),
             toaerrs=np.asarray(toaerrs, dtype=np.float64),
             residuals=np.asarray(obj.residuals, dtype=np.float64),
             Mmat=np.asarray(obj.Mmat, dtype=np.float64),
-            backend_flags=np.asarray(getattr(obj, "backend_flags", None)),
+            backend_flags=None if bflags is None else np.asarray(bflags),
         )
 
 

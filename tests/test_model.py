@@ -127,3 +127,41 @@ def test_per_psr_tspan_bases():
     # the shared-basis CURN combination is rejected, not silently wrong
     with pytest.raises(ValueError):
         initialize_pta(psrs, noise, inc_cp=True, per_psr_tspan=True)
+
+
+def test_from_object_duck_typed_pickle_surface(tmp_path):
+    """The enterprise-pickle converter: with and without backend_flags
+    / toaerrs (missing flags must fall back to the per-TOA default,
+    not a 0-d None array)."""
+    import pickle
+    import types
+
+    from fastfp_amd.data import PulsarData
+
+    src = make_synthetic_pta(npsr=1, ntoa=40, ntm=3, seed=7)[0]
+    full = types.SimpleNamespace(
+        name=src.name, toas=src.toas, toaerrs=src.toaerrs,
+        residuals=src.residuals, Mmat=src.Mmat,
+        backend_flags=src.backend_flags,
+    )
+    minimal = types.SimpleNamespace(  # no backend_flags, no toaerrs
+        name=src.name, toas=src.toas, residuals=src.residuals,
+        Mmat=src.Mmat,
+    )
+    pkl = tmp_path / "psrs.pkl"
+    with open(pkl, "wb") as f:
+        pickle.dump([full, minimal], f)
+    a, b = load_pulsars(str(pkl))
+    assert isinstance(a, PulsarData) and isinstance(b, PulsarData)
+    np.testing.assert_array_equal(a.backend_flags, src.backend_flags)
+    # missing flags -> one default backend per TOA (1-d, right length)
+    assert b.backend_flags.shape == (b.ntoa,)
+    assert set(np.unique(b.backend_flags)) == {"backend"}
+    np.testing.assert_array_equal(b.toaerrs, np.full(b.ntoa, 1e-6))
+    # the converted pulsar must drive the full model build
+    noise = {"gw_gamma": 13.0 / 3.0, "gw_log10_A": np.log10(2e-15),
+             f"{b.name}_red_noise_gamma": 4.0,
+             f"{b.name}_red_noise_log10_A": -14.5}
+    pta = initialize_pta([b], noise, inc_cp=True, rn_comps=3, gwb_comps=2)
+    TNTs, Nvecs, Ts = get_mats_nmfp(pta, noise)
+    assert Ts[0].shape[0] == b.ntoa
