@@ -62,16 +62,20 @@ def test_bench_torchrun_ws2_gloo():
     assert j["config"]["spectrum_shape"] == [8, 8]  # gathered over ranks
 
 
-def test_bench_torchrun_ws4_gloo():
-    """4 ranks — the driver's N=4 scaling launch shape.  Exercises the
-    >2-rank collective path (size exchange + padded all-gather over 4
-    shards, MAX-over-ranks timing) that the 4- and 8-GPU round-end runs
-    hit."""
+import pytest
+
+
+@pytest.mark.parametrize("world,port", [(4, 29878), (8, 29879)])
+def test_bench_torchrun_multirank_gloo(world, port):
+    """4 and 8 ranks — the driver's N=4/N=8 scaling launch shapes.
+    Exercises the >2-rank collective path (size exchange + padded
+    all-gather over N shards, MAX-over-ranks timing) that the 4- and
+    8-GPU round-end runs hit."""
     out = subprocess.run(
         [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
-         "--nproc-per-node", "4", "--master-addr", "127.0.0.1",
-         "--master-port", "29878", os.path.join(REPO, "bench.py"),
-         "--gpus", "4", "--npsr", "2", "--ntoa", "120", "--ntm", "4",
+         "--nproc-per-node", str(world), "--master-addr", "127.0.0.1",
+         "--master-port", str(port), os.path.join(REPO, "bench.py"),
+         "--gpus", str(world), "--npsr", "2", "--ntoa", "120", "--ntm", "4",
          "--rn-comps", "3", "--gwb-comps", "3", "--freqs", "8",
          "--draws-per-step", "3", "--steps", "1", "--warmup", "0",
          "--device", "cpu"],
@@ -81,6 +85,6 @@ def test_bench_torchrun_ws4_gloo():
     line = [l for l in out.stdout.strip().splitlines()
             if l.startswith("{")][-1]
     j = json.loads(line)
-    assert j["n_gpus"] == 4
-    assert j["config"]["global_batch"] == 12  # 4 ranks x 3 draws
-    assert j["config"]["spectrum_shape"] == [12, 8]  # gathered over ranks
+    assert j["n_gpus"] == world
+    assert j["config"]["global_batch"] == 3 * world
+    assert j["config"]["spectrum_shape"] == [3 * world, 8]
