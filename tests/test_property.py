@@ -1,0 +1,165 @@
+"""Hypothesis property tests of the numerical core (CPU).
+
+Randomized, shrinking counterexamples — complements the fixed-seed
+fuzz in ``test_engine.py`` and the GPU fuzz campaign
+(``tools/fuzz_gpu.py``) with property-style coverage of:
+
+- the Sherman–Morrison block-noise inverse vs the dense oracle at
+  arbitrary partitions and variance/ECORR scales,
+- the Woodbury ``get_xCy`` vs the dense ``C^{-1}`` solve,
+- the eager engine's closed-form 2x2 reduction vs the dense Fp oracle.
+
+Runtimes are bounded (small systems, capped example counts).
+"""
+
+import numpy as np
+import pytest
+import torch
+from hypothesis import given, settings, HealthCheck
+from hypothesis import strategies as st
+
+from tests.oracle import dense_xCy
+
+# derandomize: the example stream is fixed so CI/driver runs are
+# deterministic (a fresh random stream could surface a new
+# conditioning corner mid-release; exploration happens locally by
+# removing the flag)
+SET = dict(
+    max_examples=40,
+    deadline=None,
+    derandomize=True,
+    suppress_health_check=[HealthCheck.too_slow],
+)
+
+
+# ----------------------------------------------------------------------
+# BlockNoise: Sherman–Morrison vs dense block inverse
+# ----------------------------------------------------------------------
+@settings(**SET)
+@given(
+    sizes=st.lists(st.integers(1, 9), min_size=1, max_size=6),
+    log_nvec=st.floats(-14.0, 2.0),
+    log_e2=st.floats(-16.0, 0.0),
+    seed=st.integers(0, 2**31 - 1),
+)
+def test_blocknoise_sherman_morrison_property(sizes, log_nvec, log_e2, seed):
+    """N^{-1} X from the O(n) Sherman–Morrison path equals the dense
+    per-block inverse for ANY partition and variance/ECORR scales."""
+    from fastfp_amd.blocknoise import BlockNoise
+
+    rng = np.random.default_rng(seed)
+    n = sum(sizes)
+    nvec = 10.0 ** log_nvec * (0.5 + rng.random(n))
+    e2 = 10.0 ** log_e2
+
+    bn = BlockNoise.__new__(BlockNoise)
+    bn.perm = np.arange(n)
+    bn.sizes = np.asarray(sizes, dtype=np.int64)
+    bn.offsets = np.concatenate(([0], np.cumsum(bn.sizes)[:-1]))
+    # singleton blocks carry no ECORR (mirrors the real partitioner)
+    bn.ecorr2 = np.asarray(
+        [e2 if s > 1 else 0.0 for s in sizes], dtype=np.float64
+    )
+    bn.nvec = nvec
+    bn.max_block = int(bn.sizes.max())
+    bn._factor()
+
+    X = rng.standard_normal((n, 3))
+    got = bn.solve(X)
+    dense = bn.dense()
+    # both checks are conditioning-aware: at e2/nvec ~ 1e14 ANY
+    # backward-stable solve (including the dense numpy reference, and
+    # the residual of an exact-form inverse applied in fp64) carries
+    # ~cond*eps relative round-off — a fixed tolerance would test the
+    # generator's conditioning, not the Sherman–Morrison identity.
+    # Measured slope is ~1e-16*cond for both quantities.
+    cond = 1.0 + float(bn.ecorr2.max() * (1.0 / nvec).sum())
+    resid = np.abs(dense @ got - X).max() / (np.abs(X).max() + 1e-300)
+    assert resid < max(1e-12, 3e-15 * cond)
+    want = np.linalg.solve(dense, X)
+    scale = np.abs(want).max() + 1e-300
+    assert np.abs(got - want).max() / scale < max(1e-12, 3e-15 * cond)
+
+    # logdet agrees with the dense slogdet (same conditioning caveat)
+    sign, ld = np.linalg.slogdet(dense)
+    assert sign > 0
+    np.testing.assert_allclose(
+        bn.logdet, ld, rtol=max(1e-11, 3e-15 * cond), atol=1e-10
+    )
+
+
+# ----------------------------------------------------------------------
+# get_xCy: Woodbury vs dense C^{-1}
+# ----------------------------------------------------------------------
+@settings(**SET)
+@given(
+    ntoa=st.integers(4, 24),
+    m=st.integers(1, 8),
+    log_phi=st.floats(-12.0, 6.0),
+    seed=st.integers(0, 2**31 - 1),
+)
+def test_get_xcy_woodbury_property(ntoa, m, log_phi, seed):
+    from fastfp_amd.xcy import get_xCy
+
+    rng = np.random.default_rng(seed)
+    Nvec = 0.5 + rng.random(ntoa)
+    T = rng.standard_normal((ntoa, m))
+    phi = 10.0 ** log_phi * (0.5 + rng.random(m))
+    sigma = T.T @ (T / Nvec[:, None]) + np.diag(1.0 / phi)
+    x = rng.standard_normal(ntoa)
+    y = rng.standard_normal(ntoa)
+    got = get_xCy(Nvec, T, sigma, x, y)
+    want = dense_xCy(Nvec, T, phi, x, y)
+    scale = max(abs(want), abs(float(x @ (y / Nvec)))) + 1e-300
+    assert abs(got - want) / scale < 1e-8
+
+
+# ----------------------------------------------------------------------
+# Eager engine vs the dense Fp oracle at random tiny models
+# ----------------------------------------------------------------------
+@settings(max_examples=15, deadline=None, derandomize=True,
+          suppress_health_check=[HealthCheck.too_slow])
+@given(
+    npsr=st.integers(1, 3),
+    nf=st.integers(1, 4),
+    ncomps=st.integers(2, 5),
+    seed=st.integers(0, 2**31 - 1),
+)
+def test_engine_matches_dense_oracle_property(npsr, nf, ncomps, seed):
+    """Eager engine vs the dense C^{-1} oracle at random tiny models.
+
+    Priors are MODERATE (the dense np.linalg.inv oracle cannot
+    represent the production 1e40 improper-flat tm prior — that regime
+    is covered by test_engine.test_fp_invariant_to_tm_prior_scale's
+    projection-limit test instead)."""
+    from fastfp_amd import FpEngine, make_synthetic_pta
+    from fastfp_amd.bases import (
+        create_freqarray,
+        fourier_basis,
+        timing_model_basis_svd,
+    )
+    from fastfp_amd.noise import white_noise_nvec
+    from tests.oracle import dense_fp_sweep
+
+    psrs = make_synthetic_pta(npsr=npsr, ntoa=50, ntm=3, seed=seed % 1000)
+    rng = np.random.default_rng(seed // 1000 + 7)
+    Nvecs, Ts, phis = [], [], []
+    for p in psrs:
+        Nvecs.append(white_noise_nvec(p))
+        U = timing_model_basis_svd(p.Mmat)
+        Fb = fourier_basis(p.toas, create_freqarray(p.Tspan, ncomps))
+        T = np.concatenate([U, Fb], axis=1)
+        Ts.append(T)
+        phis.append(
+            np.concatenate([
+                np.full(U.shape[1], 1e5) * 1e-12,
+                rng.uniform(0.2, 5.0, Fb.shape[1]) * 1e-12,
+            ])
+        )
+    freqs = np.linspace(3e-9, 6e-8, nf)
+
+    eng = FpEngine(psrs, Nvecs, Ts, device="cpu")
+    eng.precompute(freqs)
+    got = eng.sweep(phiinvs=[1.0 / p for p in phis]).numpy()
+    want = dense_fp_sweep(psrs, Nvecs, Ts, phis, freqs)
+    np.testing.assert_allclose(got, want, rtol=1e-6, atol=1e-9)
