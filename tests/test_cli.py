@@ -118,3 +118,27 @@ def test_run_nmfp_no_checkpoint(inputs):
     b = np.load(os.path.join(outdir2, "nm.npy"))
     np.testing.assert_array_equal(a, b)
     assert not os.path.exists(os.path.join(outdir2, ".nm.batches"))
+
+
+def test_run_fp_ecorr_kernel(inputs, tmp_path):
+    """Fp CLI with --ecorr_kernel: ECORR as block-diagonal white noise
+    (the reference's unsupported case) end to end through the CLI."""
+    tmp, psrfile, noisefile, _ = inputs
+    with open(noisefile) as f:
+        noise = json.load(f)
+    psrs_noise = dict(noise)
+    # per-backend ECORR keys for both synthetic backends of each pulsar
+    for name in {k.rsplit("_red_noise", 1)[0] for k in noise
+                 if "_red_noise_gamma" in k}:
+        for b in ("BE_A", "BE_B"):
+            psrs_noise[f"{name}_basis_ecorr_{b}_log10_ecorr"] = -6.5
+    nf2 = str(tmp_path / "noise_ec.json")
+    with open(nf2, "w") as f:
+        json.dump(psrs_noise, f)
+    out = str(tmp_path / "fp_ec")
+    run_fp.main(psrfile, nf2, out, nfreqs=4, rn_comps=3, gwb_comps=3,
+                device="cpu", ecorr_kernel=True)
+    with open(out + ".json") as f:
+        res = json.load(f)
+    assert len(res) == 4
+    assert all(np.isfinite(v) for v in res.values())
