@@ -88,3 +88,33 @@ def test_engine_matches_arbitrary_precision_reference():
         fp_obj.calculate_Fp(f, [Nvec], [T], [sigma]) for f in freqs
     ])
     np.testing.assert_allclose(par, want, rtol=2e-9)
+
+
+@pytest.mark.gpu
+def test_hip_kernels_match_arbitrary_precision_reference():
+    """The full HIP path (sigdots + sbgemm precompute, chol_batch +
+    trsm_fp solve) against the 40-digit oracle: pins the KERNELS'
+    absolute accuracy, not just their agreement with CPU fp64."""
+    import torch
+
+    if not torch.cuda.is_available():
+        pytest.skip("needs a GPU")
+    psr = make_synthetic_pta(npsr=1, ntoa=24, ntm=3, seed=5,
+                             ragged=False)[0]
+    Nvec = white_noise_nvec(psr)
+    U = timing_model_basis_svd(psr.Mmat)
+    Fb = fourier_basis(psr.toas, create_freqarray(psr.Tspan, 3))
+    T = np.concatenate([U, Fb], axis=1)
+    rng = np.random.default_rng(3)
+    phi = np.concatenate([
+        np.full(U.shape[1], 1e5) * 1e-12,
+        rng.uniform(0.3, 3.0, Fb.shape[1]) * 1e-12,
+    ])
+    freqs = np.array([4.6e-9, 1.13e-8, 2.71e-8])
+
+    want = _mp_fp(psr, Nvec, T, phi, freqs)
+
+    eng = FpEngine([psr], [Nvec], [T], device="cuda:0")
+    eng.precompute(freqs)
+    got = eng.sweep(phiinvs=[1.0 / phi]).cpu().numpy()
+    np.testing.assert_allclose(got, want, rtol=2e-9)
