@@ -163,3 +163,75 @@ def test_engine_matches_dense_oracle_property(npsr, nf, ncomps, seed):
     got = eng.sweep(phiinvs=[1.0 / p for p in phis]).numpy()
     want = dense_fp_sweep(psrs, Nvecs, Ts, phis, freqs)
     np.testing.assert_allclose(got, want, rtol=1e-6, atol=1e-9)
+
+
+# ----------------------------------------------------------------------
+# batch_phiinv: the fused homogeneous path equals per-container phi
+# ----------------------------------------------------------------------
+@settings(**SET)
+@given(
+    npsr=st.integers(1, 4),
+    D=st.integers(1, 5),
+    rn=st.integers(2, 6),
+    add_curn=st.booleans(),
+    seed=st.integers(0, 2**31 - 1),
+)
+def test_batch_phiinv_homogeneous_matches_per_container(
+    npsr, D, rn, add_curn, seed
+):
+    from fastfp_amd import initialize_pta, make_synthetic_pta
+    from fastfp_amd.noise import batch_phiinv, check_batch_homogeneous
+
+    psrs = make_synthetic_pta(npsr=npsr, ntoa=30, ntm=2, seed=seed % 997)
+    noise = {"gw_gamma": 13.0 / 3.0, "gw_log10_A": float(np.log10(2e-15))}
+    for p in psrs:
+        noise[f"{p.name}_red_noise_gamma"] = 4.0
+        noise[f"{p.name}_red_noise_log10_A"] = -14.5
+    pta = initialize_pta(psrs, noise, inc_cp=add_curn, rn_comps=rn,
+                         gwb_comps=min(rn, 2))
+    rng = np.random.default_rng(seed // 997)
+    pars = {}
+    for n in pta.params:
+        pars[n] = torch.as_tensor(
+            rng.uniform(1.5, 6.5, D) if n.endswith("gamma")
+            else rng.uniform(-16.5, -13.5, D),
+            dtype=torch.float64,
+        )
+    assert check_batch_homogeneous(pta.rn_containers)
+    fused = batch_phiinv(pta.rn_containers, pars, homogeneous=True)
+    for cont, f in zip(pta.rn_containers, fused):
+        ref = cont.get_phiinv(pars)
+        ref = ref[None, :] if ref.dim() == 1 else ref
+        np.testing.assert_allclose(f.numpy(), ref.numpy(), rtol=1e-14)
+
+
+# ----------------------------------------------------------------------
+# quantization bucketing invariants
+# ----------------------------------------------------------------------
+@settings(**SET)
+@given(
+    n=st.integers(1, 60),
+    dt_days=st.floats(0.1, 30.0),
+    nmin=st.integers(1, 4),
+    seed=st.integers(0, 2**31 - 1),
+)
+def test_quantization_matrix_invariants(n, dt_days, nmin, seed):
+    """U columns are disjoint 0/1 indicators; every column has >= nmin
+    TOAs; TOAs within one bucket span < dt from the bucket's first."""
+    from fastfp_amd.bases import create_quantization_matrix
+    from fastfp_amd.constants import day
+
+    rng = np.random.default_rng(seed)
+    toas = np.sort(rng.uniform(0, 3e8, n))
+    dt = dt_days * day
+    U, w = create_quantization_matrix(toas, dt=dt, nmin=nmin)
+    assert U.shape == (n, len(w))
+    assert set(np.unique(U)) <= {0.0, 1.0}
+    # each TOA is in at most one kept bucket
+    assert (U.sum(axis=1) <= 1.0 + 1e-15).all()
+    for j in range(U.shape[1]):
+        members = np.nonzero(U[:, j])[0]
+        assert len(members) >= nmin
+        tb = toas[members]
+        assert tb.max() - tb.min() < dt * len(members)  # chained buckets
+        assert w[j] == 1.0
