@@ -142,3 +142,34 @@ def test_run_fp_ecorr_kernel(inputs, tmp_path):
         res = json.load(f)
     assert len(res) == 4
     assert all(np.isfinite(v) for v in res.values())
+
+
+def test_run_fp_torchrun_ws2_gloo(inputs, tmp_path):
+    """The Fp CLI under the production multi-GPU launch shape (2 ranks,
+    gloo on CPU): frequency sharding + all-gather, rank-0 JSON output
+    identical in contract to the single-process run."""
+    import subprocess
+    import sys
+
+    tmp, psrfile, noisefile, _ = inputs
+    repo = os.path.abspath(os.path.join(os.path.dirname(__file__), ".."))
+    out = str(tmp_path / "fp_ws2")
+    r = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+         "--master-port", "29881", "-m", "fastfp_amd.cli.run_fp",
+         psrfile, noisefile, out, "--nfreqs", "5", "--rn_comps", "3",
+         "--gwb_comps", "3", "--device", "cpu"],
+        cwd=repo, capture_output=True, text=True, timeout=600,
+    )
+    assert r.returncode == 0, r.stderr[-2000:]
+    with open(out + ".json") as f:
+        res = json.load(f)
+    assert len(res) == 5
+    # bitwise contract: sharded == single process (fp64 determinism)
+    out1 = str(tmp_path / "fp_ws1")
+    run_fp.main(psrfile, noisefile, out1, nfreqs=5, rn_comps=3,
+                gwb_comps=3, device="cpu")
+    with open(out1 + ".json") as f:
+        res1 = json.load(f)
+    assert res == res1

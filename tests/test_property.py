@@ -235,3 +235,45 @@ def test_quantization_matrix_invariants(n, dt_days, nmin, seed):
         tb = toas[members]
         assert tb.max() - tb.min() < dt * len(members)  # chained buckets
         assert w[j] == 1.0
+
+
+# ----------------------------------------------------------------------
+# sweep is bitwise-invariant to the draw_chunk partition
+# ----------------------------------------------------------------------
+@settings(max_examples=20, deadline=None, derandomize=True,
+          suppress_health_check=[HealthCheck.too_slow])
+@given(
+    D=st.integers(1, 9),
+    chunk=st.integers(1, 12),
+    seed=st.integers(0, 2**31 - 1),
+)
+def test_sweep_bitwise_invariant_to_chunking_property(D, chunk, seed):
+    """Chunking the draw axis must not change a single bit: the
+    per-draw computations are independent and reductions have fixed
+    order (docs/DESIGN.md §3)."""
+    from fastfp_amd import FpEngine, initialize_pta, make_synthetic_pta
+    from fastfp_amd.model import get_mats_nmfp
+    from fastfp_amd.noise import batch_phiinv
+
+    psrs = make_synthetic_pta(npsr=2, ntoa=40, ntm=2, seed=seed % 991)
+    noise = {"gw_gamma": 13.0 / 3.0, "gw_log10_A": float(np.log10(2e-15))}
+    for p in psrs:
+        noise[f"{p.name}_red_noise_gamma"] = 4.0
+        noise[f"{p.name}_red_noise_log10_A"] = -14.5
+    pta = initialize_pta(psrs, noise, inc_cp=True, rn_comps=3, gwb_comps=2)
+    _, Nvecs, Ts = get_mats_nmfp(pta, noise)
+
+    rng = np.random.default_rng(seed // 991)
+    pars = {
+        n: torch.as_tensor(
+            rng.uniform(2, 6, D) if n.endswith("gamma")
+            else rng.uniform(-16, -14, D), dtype=torch.float64)
+        for n in pta.params
+    }
+    phiinvs = batch_phiinv(pta.rn_containers, pars)
+    phiinvs = [p[None, :] if p.dim() == 1 else p for p in phiinvs]
+    eng = FpEngine(psrs, Nvecs, Ts, device="cpu")
+    eng.precompute(np.linspace(4e-9, 5e-8, 3))
+    a = eng.sweep(phiinvs=phiinvs, draw_chunk=chunk).numpy()
+    b = eng.sweep(phiinvs=phiinvs, draw_chunk=D).numpy()
+    np.testing.assert_array_equal(a, b)
