@@ -196,6 +196,14 @@ def main():
         device=torch.device(args.device) if args.device else None
     )
     on_gpu = device.type == "cuda"
+    if world > 1:
+        # the compression precompute is pinned to CPU LAPACK; with N
+        # ranks each defaulting to all cores, setup would thrash on
+        # oversubscribed threads.  Timed-region work is all on-GPU, so
+        # this only protects the (untimed) setup wall time.
+        import os as _os
+
+        torch.set_num_threads(max(1, (_os.cpu_count() or world) // world))
 
     pta, eng, pool, active = build_problem(args, device, rank)
     F = args.freqs
