@@ -47,6 +47,12 @@ def main(
     rank, world, dev = init_distributed(
         device=torch.device(device) if device else None
     )
+    if world > 1:
+        # CPU-pinned precompute under N concurrent ranks: avoid
+        # oversubscribed LAPACK threading (see bench.py)
+        import os
+
+        torch.set_num_threads(max(1, (os.cpu_count() or world) // world))
     logger.info(f"fastfp_amd backend device {dev} (rank {rank}/{world})")
 
     psrs = load_pulsars(psrfile)

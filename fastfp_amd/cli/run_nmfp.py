@@ -91,6 +91,10 @@ def main(
     rank, world, dev = init_distributed(
         device=torch.device(device) if device else None
     )
+    if world > 1:
+        # CPU-pinned precompute under N concurrent ranks: avoid
+        # oversubscribed LAPACK threading (see bench.py)
+        torch.set_num_threads(max(1, (os.cpu_count() or world) // world))
     logger.info(f"fastfp_amd backend device {dev} (rank {rank}/{world})")
     logger.info(f"number of CW frequencies: {ncwfreqs}")
     logger.info(f"number of samples: {nsamples}")
