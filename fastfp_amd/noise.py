@@ -260,6 +260,21 @@ class RNContainer:
             if inc_tm
             else torch.zeros(0, dtype=torch.float64, device=self.Ffreqs.device)
         )
+        # parity attribute: the reference binds the matching phi-variant
+        # method at init (``/root/reference/fastfp/nmfp.py:185-199``);
+        # here every variant routes through the one vectorized assembly,
+        # but ``cont.phi_fn(pars)`` keeps working for reference users.
+        variants = {
+            (True, False, False): self.get_phi_tm_rn,
+            (True, False, True): self.get_phi_tm_rn_curn,
+            (True, True, False): self.get_phi_tm_ecorr_rn,
+            (True, True, True): self.get_phi_tm_ecorr_rn_curn,
+            (False, False, False): self.get_phi_rn,
+            (False, False, True): self.get_phi_rn_curn,
+            (False, True, False): self.get_phi_ecorr_rn,
+            (False, True, True): self.get_phi_ecorr_rn_curn,
+        }
+        self.phi_fn = variants[(inc_tm, gp_ecorr, add_curn)]
 
     def to(self, device):
         self.Ffreqs = self.Ffreqs.to(device)

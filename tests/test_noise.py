@@ -205,3 +205,29 @@ def test_white_noise_no_selection(psr):
     nv = white_noise_nvec(psr, noise, simple_wn=False, select="none")
     want = 1.2**2 * (psr.toaerrs**2 + 10.0 ** (2 * -6.5))
     np.testing.assert_allclose(nv, want, rtol=1e-12)
+
+
+def test_phi_fn_bound_variant(psr):
+    """Reference parity: RN_container binds the matching phi-variant
+    method at init (/root/reference/fastfp/nmfp.py:185-199) — users may
+    call ``cont.phi_fn(pars)`` directly."""
+    from fastfp_amd.noise import CURNContainer, RNContainer
+    from fastfp_amd.bases import create_freqarray
+
+    curn = CURNContainer(create_freqarray(psr.Tspan, 2))
+    pars = {
+        f"{psr.name}_red_noise_gamma": 4.0,
+        f"{psr.name}_red_noise_log10_A": -14.5,
+        "gw_gamma": 13.0 / 3.0,
+        "gw_log10_A": -14.8,
+    }
+    plain = RNContainer(psr, ncomps=3)
+    assert plain.phi_fn == plain.get_phi_tm_rn
+    np.testing.assert_array_equal(
+        plain.phi_fn(pars).numpy(), plain.update_phi(pars).numpy()
+    )
+    withc = RNContainer(psr, ncomps=3, add_curn=True, curn_container=curn)
+    assert withc.phi_fn == withc.get_phi_tm_rn_curn
+    np.testing.assert_array_equal(
+        withc.phi_fn(pars).numpy(), withc.update_phi(pars).numpy()
+    )
