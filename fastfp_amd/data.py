@@ -50,6 +50,10 @@ class PulsarData:
     residuals: np.ndarray
     Mmat: np.ndarray
     backend_flags: np.ndarray = field(default=None)
+    #: unit 3-vector pointing from the SSB to the pulsar (enterprise's
+    #: ``Pulsar.pos``); needed by the sky-coherent Fe statistic.  None
+    #: for data sources that carry no astrometry.
+    pos: np.ndarray = field(default=None)
 
     def __post_init__(self):
         self.toas = np.asarray(self.toas, dtype=np.float64)
@@ -60,6 +64,9 @@ class PulsarData:
             self.backend_flags = np.array(["backend"] * self.ntoa, dtype=object)
         else:
             self.backend_flags = np.asarray(self.backend_flags)
+        if self.pos is not None:
+            self.pos = np.asarray(self.pos, dtype=np.float64)
+            self.pos = self.pos / np.linalg.norm(self.pos)
 
     # ------------------------------------------------------------------
     @property
@@ -76,8 +83,7 @@ class PulsarData:
 
     # ------------------------------------------------------------------
     def save_npz(self, path: str) -> None:
-        np.savez_compressed(
-            path,
+        arrs = dict(
             name=np.asarray(self.name),
             toas=self.toas,
             toaerrs=self.toaerrs,
@@ -85,6 +91,9 @@ class PulsarData:
             Mmat=self.Mmat,
             backend_flags=np.asarray(self.backend_flags, dtype=str),
         )
+        if self.pos is not None:
+            arrs["pos"] = self.pos
+        np.savez_compressed(path, **arrs)
 
     @classmethod
     def load_npz(cls, path: str) -> "PulsarData":
@@ -96,6 +105,7 @@ class PulsarData:
             residuals=z["residuals"],
             Mmat=z["Mmat"],
             backend_flags=z["backend_flags"].astype(object),
+            pos=z["pos"] if "pos" in z.files else None,
         )
 
     def save_feather(self, path: str) -> None:
@@ -116,9 +126,10 @@ class PulsarData:
         }
         for j in range(self.ntm):
             cols[f"Mmat_{j}"] = self.Mmat[:, j]
-        table = pa.table(cols).replace_schema_metadata(
-            {"fastfp_amd.name": self.name, "fastfp_amd.ntm": str(self.ntm)}
-        )
+        meta = {"fastfp_amd.name": self.name, "fastfp_amd.ntm": str(self.ntm)}
+        if self.pos is not None:
+            meta["fastfp_amd.pos"] = ",".join(repr(float(v)) for v in self.pos)
+        table = pa.table(cols).replace_schema_metadata(meta)
         # the Arrow IPC file format IS Feather V2
         with paipc.new_file(path, table.schema) as w:
             w.write_table(table)
@@ -132,6 +143,11 @@ class PulsarData:
         meta = table.schema.metadata or {}
         name = meta.get(b"fastfp_amd.name", b"unknown").decode()
         ntm = int(meta.get(b"fastfp_amd.ntm", b"0"))
+        pos_s = meta.get(b"fastfp_amd.pos", None)
+        pos = (
+            np.array([float(v) for v in pos_s.decode().split(",")])
+            if pos_s else None
+        )
         cols = {c: table[c].to_numpy() for c in table.column_names}
         Mmat = np.stack(
             [cols[f"Mmat_{j}"] for j in range(ntm)], axis=1
@@ -143,6 +159,7 @@ class PulsarData:
             residuals=cols["residuals"],
             Mmat=Mmat,
             backend_flags=cols["backend_flags"].astype(object),
+            pos=pos,
         )
 
     @classmethod
@@ -157,6 +174,7 @@ class PulsarData:
         # 0-d object array, which would bypass the per-TOA default in
         # __post_init__ and break backend masking downstream)
         bflags = getattr(obj, "backend_flags", None)
+        pos = getattr(obj, "pos", None)
         return cls(
             name=str(obj.name),
             toas=np.asarray(obj.toas, dtype=np.float64),
@@ -164,6 +182,7 @@ class PulsarData:
             residuals=np.asarray(obj.residuals, dtype=np.float64),
             Mmat=np.asarray(obj.Mmat, dtype=np.float64),
             backend_flags=None if bflags is None else np.asarray(bflags),
+            pos=None if pos is None else np.asarray(pos, dtype=np.float64),
         )
 
 
@@ -211,6 +230,7 @@ def load_pulsars(path: str) -> list:
                     residuals=z[f"residuals_{i}"],
                     Mmat=z[f"Mmat_{i}"],
                     backend_flags=z[f"backend_flags_{i}"].astype(object),
+                    pos=z[f"pos_{i}"] if f"pos_{i}" in z.files else None,
                 )
             )
         return out
@@ -227,6 +247,8 @@ def save_pulsars(psrs: list, path: str) -> None:
         arrs[f"residuals_{i}"] = p.residuals
         arrs[f"Mmat_{i}"] = p.Mmat
         arrs[f"backend_flags_{i}"] = np.asarray(p.backend_flags, dtype=str)
+        if p.pos is not None:
+            arrs[f"pos_{i}"] = p.pos
     np.savez_compressed(path, **arrs)
 
 
@@ -319,6 +341,8 @@ def make_pulsar(
         [backends[int(e) % len(backends)] for e in epoch_of], dtype=object
     )
     Mmat = _design_matrix(toas, ntm, rng)
+    # isotropic sky position (unit vector) for the sky-coherent Fe path
+    v = rng.normal(size=3)
     return PulsarData(
         name=name,
         toas=toas,
@@ -326,6 +350,7 @@ def make_pulsar(
         residuals=resid,
         Mmat=Mmat,
         backend_flags=bflags,
+        pos=v / np.linalg.norm(v),
     )
 
 
