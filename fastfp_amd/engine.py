@@ -816,6 +816,43 @@ class FpEngine:
             fp[lo:hi] += pp.sum(dim=0)
         return fp[0] if not batched else fp
 
+    def sweep_products(self, sigmas=None, phiinvs=None) -> torch.Tensor:
+        """Per-pulsar corrected inner products at fixed noise:
+        returns (P, 5, F) with rows [s|s, c|c, s|c, s|r, c|r], each
+        ``(x|y) = x^T C_p^{-1} y`` over the precomputed frequency grid.
+
+        These are exactly the quantities the Fp reduction consumes
+        before its per-pulsar 2x2 solve; exposing them lets sky-
+        coherent statistics (the Fe assembly, ``fastfp_amd.festat``)
+        reuse the engine's precompute + one m-dim solve per pulsar and
+        pay only O(P) per additional sky location.  Runs on the
+        engine's device (torch Cholesky/solve — a one-off per-pulsar
+        factor, not the draw-batched hot path)."""
+        assert self.freqs is not None, "call precompute(freqs) first"
+        F = self.freqs.shape[0]
+        out = torch.empty((len(self.blocks), 5, F), dtype=torch.float64,
+                          device=self.device)
+        for i, blk in enumerate(self.blocks):
+            if sigmas is not None:
+                sigma = _t64(sigmas[i], self.device)
+                sigma = sigma if sigma.dim() == 2 else sigma[0]
+            else:
+                pinv = _t64(phiinvs[i], self.device).reshape(-1)
+                sigma = blk.TNT + torch.diag(pinv)
+            m = blk.m
+            L = torch.linalg.cholesky(sigma)
+            W = torch.linalg.solve_triangular(
+                L, blk.RHS[:m, :], upper=False
+            )
+            wu = W[:, -1]
+            Ws, Wc = W[:, 0:-1:2], W[:, 1:-1:2]
+            out[i, 0] = blk.sNs[0] - (Ws * Ws).sum(0)
+            out[i, 1] = blk.sNs[1] - (Wc * Wc).sum(0)
+            out[i, 2] = blk.sNs[2] - (Ws * Wc).sum(0)
+            out[i, 3] = blk.sNr[0] - Ws.transpose(0, 1) @ wu
+            out[i, 4] = blk.sNr[1] - Wc.transpose(0, 1) @ wu
+        return out
+
     def _accum_eager(self, blk: PulsarBlock, sigma, fp_out):
         """Eager per-pulsar accumulation: Cholesky + TRSM + fused 2x2."""
         self._accum_eager_mats(sigma, blk.RHS, blk.sNs, blk.sNr, fp_out, 1.0)

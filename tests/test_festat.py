@@ -1,0 +1,214 @@
+"""Fe-statistic tests: antenna-pattern properties, dense-oracle
+agreement, engine-sweep consistency, and coherent CW detection with
+sky localization.  (The Fe statistic is the reference's open to-do,
+``/root/reference/README.md:23`` — implemented natively here.)"""
+
+import math
+
+import numpy as np
+import pytest
+
+from fastfp_amd import FastFp, initialize_pta, make_synthetic_pta
+from fastfp_amd.festat import FastFe, _assemble_fe, gw_antenna_pattern
+from fastfp_amd.model import get_mats_fp
+from tests.oracle import dense_xCy
+
+
+def _pta(npsr=4, ntoa=90, seed=0, **mk):
+    psrs = make_synthetic_pta(npsr=npsr, ntoa=ntoa, ntm=3, seed=seed, **mk)
+    noise = {"gw_gamma": 13.0 / 3.0, "gw_log10_A": float(np.log10(2e-15))}
+    for p in psrs:
+        noise[f"{p.name}_red_noise_gamma"] = 4.0
+        noise[f"{p.name}_red_noise_log10_A"] = -14.5
+    pta = initialize_pta(psrs, noise, inc_cp=True, rn_comps=3, gwb_comps=3)
+    return psrs, noise, pta
+
+
+# ----------------------------------------------------------------------
+# antenna patterns
+# ----------------------------------------------------------------------
+def test_antenna_pattern_tensor_form():
+    """F+/Fx match the polarization-tensor definition
+    F_A = 1/2 p.e_A.p / (1 + omhat.p) with e+ = mm - nn, ex = mn + nm."""
+    rng = np.random.default_rng(2)
+    for _ in range(20):
+        gwtheta = rng.uniform(0.1, np.pi - 0.1)
+        gwphi = rng.uniform(0, 2 * np.pi)
+        p = rng.normal(size=3)
+        p /= np.linalg.norm(p)
+        st, ct = math.sin(gwtheta), math.cos(gwtheta)
+        sp, cp = math.sin(gwphi), math.cos(gwphi)
+        m = np.array([sp, -cp, 0.0])
+        n = np.array([-ct * cp, -ct * sp, st])
+        om = np.array([-st * cp, -st * sp, -ct])
+        if abs(1 + om @ p) < 1e-3:
+            continue
+        eplus = np.outer(m, m) - np.outer(n, n)
+        ecross = np.outer(m, n) + np.outer(n, m)
+        want_p = 0.5 * (p @ eplus @ p) / (1 + om @ p)
+        want_x = 0.5 * (p @ ecross @ p) / (1 + om @ p)
+        fp_, fx_ = gw_antenna_pattern(p, gwtheta, gwphi)
+        np.testing.assert_allclose([fp_, fx_], [want_p, want_x],
+                                   rtol=1e-12, atol=1e-14)
+
+
+def test_antenna_pattern_polarization_invariant():
+    """Rotating the (m, n) polarization frame by psi rotates (F+, Fx)
+    by 2*psi — F+^2 + Fx^2 is frame-invariant.  Checked against an
+    explicitly rotated tensor construction."""
+    rng = np.random.default_rng(5)
+    gwtheta, gwphi = 1.1, 2.3
+    st, ct = math.sin(gwtheta), math.cos(gwtheta)
+    sp, cp = math.sin(gwphi), math.cos(gwphi)
+    m = np.array([sp, -cp, 0.0])
+    n = np.array([-ct * cp, -ct * sp, st])
+    om = np.array([-st * cp, -st * sp, -ct])
+    for _ in range(10):
+        p = rng.normal(size=3)
+        p /= np.linalg.norm(p)
+        fp0, fx0 = gw_antenna_pattern(p, gwtheta, gwphi)
+        for psi in (0.3, 1.0, 2.2):
+            mr = math.cos(psi) * m + math.sin(psi) * n
+            nr = -math.sin(psi) * m + math.cos(psi) * n
+            e_p = np.outer(mr, mr) - np.outer(nr, nr)
+            e_x = np.outer(mr, nr) + np.outer(nr, mr)
+            fpr = 0.5 * (p @ e_p @ p) / (1 + om @ p)
+            fxr = 0.5 * (p @ e_x @ p) / (1 + om @ p)
+            np.testing.assert_allclose(fpr**2 + fxr**2, fp0**2 + fx0**2,
+                                       rtol=1e-10)
+
+
+# ----------------------------------------------------------------------
+# Fe vs dense oracle
+# ----------------------------------------------------------------------
+def _moderate_setup(seed=7, npsr=4, ntoa=90):
+    """Moderate-prior (Nvecs, Ts, phis, sigmas): the dense np.linalg
+    oracle cannot represent the production 1e40 tm prior (cond ~1e52;
+    same restriction as tests/test_engine._tiny_setup)."""
+    from fastfp_amd.bases import (
+        create_freqarray,
+        fourier_basis,
+        timing_model_basis_svd,
+    )
+    from fastfp_amd.noise import white_noise_nvec
+
+    psrs = make_synthetic_pta(npsr=npsr, ntoa=ntoa, ntm=3, seed=seed)
+    rng = np.random.default_rng(seed + 50)
+    Nvecs, Ts, phis, sigmas = [], [], [], []
+    for p in psrs:
+        Nvecs.append(white_noise_nvec(p))
+        U = timing_model_basis_svd(p.Mmat)
+        Fb = fourier_basis(p.toas, create_freqarray(p.Tspan, 3))
+        T = np.concatenate([U, Fb], axis=1)
+        Ts.append(T)
+        phi = np.concatenate([
+            np.full(U.shape[1], 1e5) * 1e-12,
+            rng.uniform(0.3, 3.0, Fb.shape[1]) * 1e-12,
+        ])
+        phis.append(phi)
+        sigmas.append(T.T @ (T / Nvecs[-1][:, None]) + np.diag(1.0 / phi))
+    return psrs, Nvecs, Ts, phis, sigmas
+
+
+def test_calculate_fe_matches_dense_oracle():
+    """Single-point Fe equals the fully dense construction: build the
+    four filters A_i explicitly per pulsar, form N/M with dense C^{-1}
+    inner products, solve the 4x4."""
+    psrs, Nvecs, Ts, phis, sigmas = _moderate_setup(seed=7)
+    fgw, gwtheta, gwphi = 1.3e-8, 0.9, 4.0
+
+    fe = FastFe(psrs)
+    got = fe.calculate_Fe(fgw, gwtheta, gwphi, Nvecs, Ts, sigmas)
+
+    N = np.zeros(4)
+    M = np.zeros((4, 4))
+    for p, Nvec, T, phi in zip(psrs, Nvecs, Ts, phis):
+        fp_, fx_ = gw_antenna_pattern(p.pos, gwtheta, gwphi)
+        arg = 2 * np.pi * fgw * p.toas
+        s, c = np.sin(arg), np.cos(arg)
+        A = [fp_ * s, fp_ * c, fx_ * s, fx_ * c]
+        for i in range(4):
+            N[i] += dense_xCy(Nvec, T, phi, A[i], p.residuals)
+            for j in range(4):
+                M[i, j] += dense_xCy(Nvec, T, phi, A[i], A[j])
+    want = 0.5 * float(N @ np.linalg.solve(M, N))
+    np.testing.assert_allclose(got, want, rtol=1e-7)
+
+
+def test_sweep_matches_calculate_fe():
+    psrs, noise, pta = _pta(seed=9)
+    Nvecs, Ts, sigmas = get_mats_fp(pta, noise)
+    freqs = np.array([6e-9, 1.4e-8, 3.1e-8])
+    sky = [(0.7, 1.0), (2.1, 5.5)]
+    fe = FastFe(psrs, pta)
+    grid = fe.sweep(freqs, sky, Nvecs, Ts, sigmas, device="cpu")
+    assert grid.shape == (2, 3)
+    for k, (th, ph) in enumerate(sky):
+        for fi, f in enumerate(freqs):
+            want = fe.calculate_Fe(f, th, ph, Nvecs, Ts, sigmas)
+            np.testing.assert_allclose(grid[k, fi], want, rtol=1e-8)
+
+
+def test_missing_pos_raises():
+    psrs, noise, pta = _pta(npsr=2, seed=11)
+    psrs[1].pos = None
+    with pytest.raises(ValueError, match="sky position"):
+        FastFe(psrs)
+
+
+# ----------------------------------------------------------------------
+# physics: coherent detection + localization
+# ----------------------------------------------------------------------
+def test_fe_detects_and_localizes_injected_cw():
+    """Inject a coherent Earth-term CW (antenna-weighted filters, one
+    sky location, one frequency) into a quiet PTA: the Fe map must
+    peak at the injected frequency, and the true sky must beat a
+    far-away sky at that frequency."""
+    rng = np.random.default_rng(13)
+    psrs, noise, pta = _pta(npsr=6, ntoa=120, seed=13, toaerr=1e-7)
+    f0, th0, ph0 = 1.6e-8, 1.0, 2.0
+    amps = np.array([4e-7, -2e-7, 3e-7, 1e-7])
+    for p in psrs:
+        fp_, fx_ = gw_antenna_pattern(p.pos, th0, ph0)
+        arg = 2 * np.pi * f0 * p.toas
+        s, c = np.sin(arg), np.cos(arg)
+        p.residuals = p.residuals + (
+            amps[0] * fp_ * s + amps[1] * fp_ * c
+            + amps[2] * fx_ * s + amps[3] * fx_ * c
+        )
+    Nvecs, Ts, sigmas = get_mats_fp(pta, noise)
+    freqs = np.linspace(5e-9, 4e-8, 25)
+    fe = FastFe(psrs, pta)
+    grid = fe.sweep(freqs, [(th0, ph0)], Nvecs, Ts, sigmas, device="cpu")
+    i0 = int(np.argmin(np.abs(freqs - f0)))
+    assert int(np.argmax(grid[0])) in (i0 - 1, i0, i0 + 1)
+    others = np.delete(grid[0], [max(0, i0 - 1), i0, min(24, i0 + 1)])
+    assert grid[0].max() > 10 * np.median(others)
+    # localization: the true sky is the maximum over a coarse sky
+    # sample at the injected frequency (a 4-amplitude fit can absorb
+    # part of the signal anywhere, so wrong skies are suppressed but
+    # not zero — argmax is the robust statement)
+    others_sky = [(0.4, 0.5), (2.6, 1.4), (1.5, 5.2),
+                  (np.pi - th0, (ph0 + np.pi) % (2 * np.pi))]
+    fmap = fe.sweep(np.array([freqs[i0]]), [(th0, ph0)] + others_sky,
+                    Nvecs, Ts, sigmas, device="cpu")[:, 0]
+    assert int(np.argmax(fmap)) == 0, fmap
+    assert fmap[0] > 1.5 * np.median(fmap[1:])
+
+
+# ----------------------------------------------------------------------
+# assembly degeneracy: pseudo-inverse branch
+# ----------------------------------------------------------------------
+def test_assemble_fe_singular_sky():
+    """A sky/array with Fx = 0 for every pulsar makes M rank-2; the
+    assembly must return the finite column-space maximum instead of
+    raising."""
+    rng = np.random.default_rng(3)
+    P, F = 3, 4
+    prods = np.abs(rng.normal(2.0, 0.3, (P, 5, F)))
+    prods[:, 2] *= 0.1  # sc
+    fplus = rng.normal(size=P)
+    fcross = np.zeros(P)
+    out = _assemble_fe(prods, fplus, fcross)
+    assert out.shape == (F,)
+    assert np.isfinite(out).all()
