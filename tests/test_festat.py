@@ -212,3 +212,22 @@ def test_assemble_fe_singular_sky():
     out = _assemble_fe(prods, fplus, fcross)
     assert out.shape == (F,)
     assert np.isfinite(out).all()
+
+
+@pytest.mark.gpu
+def test_fe_sweep_gpu_matches_cpu():
+    """Fe on the HIP engine (GPU precompute kernels + on-device
+    products) vs the CPU path — validated on MI355X
+    (profiles/r02_fe_gpu_validation.log: rel err 8e-16)."""
+    import torch
+
+    if not torch.cuda.is_available():
+        pytest.skip("needs a GPU")
+    psrs, noise, pta = _pta(npsr=3, ntoa=200, seed=4)
+    Nvecs, Ts, sigmas = get_mats_fp(pta, noise)
+    freqs = np.linspace(5e-9, 4e-8, 7)
+    sky = [(0.8, 1.2), (2.0, 4.4)]
+    fe = FastFe(psrs, pta)
+    g = fe.sweep(freqs, sky, Nvecs, Ts, sigmas, device="cuda:0")
+    c = fe.sweep(freqs, sky, Nvecs, Ts, sigmas, device="cpu")
+    np.testing.assert_allclose(g, c, rtol=1e-9)
