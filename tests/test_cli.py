@@ -173,3 +173,39 @@ def test_run_fp_torchrun_ws2_gloo(inputs, tmp_path):
     with open(out1 + ".json") as f:
         res1 = json.load(f)
     assert res == res1
+
+
+def test_run_fe_json_output(inputs, tmp_path):
+    """Fe CLI: (sky x freq) map JSON with the documented keys; single
+    sky location mode too."""
+    from fastfp_amd.cli import run_fe
+
+    tmp, psrfile, noisefile, _ = inputs
+    out = str(tmp_path / "feout")
+    run_fe.main(psrfile, noisefile, out, nfreqs=4, nsky=6, rn_comps=3,
+                gwb_comps=3, device="cpu")
+    with open(out + ".json") as f:
+        res = json.load(f)
+    assert len(res["freqs"]) == 4 and len(res["sky"]) == 6
+    fe = np.asarray(res["fe"])
+    assert fe.shape == (6, 4) and np.isfinite(fe).all()
+    # single-sky mode
+    out1 = str(tmp_path / "feout1")
+    run_fe.main(psrfile, noisefile, out1, nfreqs=3, theta=1.0, phi=2.0,
+                rn_comps=3, gwb_comps=3, device="cpu")
+    with open(out1 + ".json") as f:
+        res1 = json.load(f)
+    assert np.asarray(res1["fe"]).shape == (1, 3)
+
+
+def test_fibonacci_sky_coverage():
+    from fastfp_amd.cli.run_fe import fibonacci_sky
+
+    sky = fibonacci_sky(100)
+    th = np.array([t for t, _ in sky])
+    ph = np.array([p for _, p in sky])
+    assert ((th >= 0) & (th <= np.pi)).all()
+    assert ((ph >= 0) & (ph < 2 * np.pi)).all()
+    # equal-area: z = cos(theta) should be ~uniform on [-1, 1]
+    z = np.sort(np.cos(th))
+    assert np.abs(z - np.linspace(z[0], z[-1], 100)).max() < 0.03
