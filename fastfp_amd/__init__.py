@@ -35,7 +35,7 @@ from fastfp_amd.model import (  # noqa: F401
 )
 from fastfp_amd.xcy import get_xCy  # noqa: F401
 from fastfp_amd.fpstat import FastFp, compute_Fp  # noqa: F401
-from fastfp_amd.festat import FastFe, compute_Fe  # noqa: F401
+from fastfp_amd.festat import FastFe, NMFe, compute_Fe  # noqa: F401
 from fastfp_amd.nmfp import NMFp  # noqa: F401
 from fastfp_amd.engine import FpEngine  # noqa: F401
 

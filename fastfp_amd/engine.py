@@ -817,41 +817,52 @@ class FpEngine:
         return fp[0] if not batched else fp
 
     def sweep_products(self, sigmas=None, phiinvs=None) -> torch.Tensor:
-        """Per-pulsar corrected inner products at fixed noise:
-        returns (P, 5, F) with rows [s|s, c|c, s|c, s|r, c|r], each
-        ``(x|y) = x^T C_p^{-1} y`` over the precomputed frequency grid.
+        """Per-pulsar corrected inner products
+        [s|s, c|c, s|c, s|r, c|r], each ``(x|y) = x^T C_p^{-1} y``
+        over the precomputed frequency grid.
+
+        Fixed noise ((m,) phiinvs / (m, m) sigmas): returns (P, 5, F).
+        Draw-batched ((D, m) phiinvs / (D, m, m) sigmas): returns
+        (P, D, 5, F) — the noise-marginalized Fe input.
 
         These are exactly the quantities the Fp reduction consumes
         before its per-pulsar 2x2 solve; exposing them lets sky-
         coherent statistics (the Fe assembly, ``fastfp_amd.festat``)
-        reuse the engine's precompute + one m-dim solve per pulsar and
-        pay only O(P) per additional sky location.  Runs on the
-        engine's device (torch Cholesky/solve — a one-off per-pulsar
-        factor, not the draw-batched hot path)."""
+        reuse the engine's precompute + one m-dim solve per (pulsar,
+        draw) and pay only O(P) per additional sky location.  Runs on
+        the engine's device (torch batched Cholesky/solve — setup-
+        scale work, not the draw-batched Fp hot path)."""
         assert self.freqs is not None, "call precompute(freqs) first"
         F = self.freqs.shape[0]
-        out = torch.empty((len(self.blocks), 5, F), dtype=torch.float64,
-                          device=self.device)
+        vals = sigmas if sigmas is not None else phiinvs
+        first = _t64(vals[0], self.device)
+        batched = first.dim() == (3 if sigmas is not None else 2)
+        D = first.shape[0] if batched else 1
+        out = torch.empty((len(self.blocks), D, 5, F),
+                          dtype=torch.float64, device=self.device)
         for i, blk in enumerate(self.blocks):
             if sigmas is not None:
                 sigma = _t64(sigmas[i], self.device)
-                sigma = sigma if sigma.dim() == 2 else sigma[0]
+                sigma = sigma[None] if sigma.dim() == 2 else sigma
             else:
-                pinv = _t64(phiinvs[i], self.device).reshape(-1)
-                sigma = blk.TNT + torch.diag(pinv)
+                pinv = _t64(phiinvs[i], self.device)
+                pinv = pinv[None, :] if pinv.dim() == 1 else pinv
+                sigma = blk.TNT[None] + torch.diag_embed(pinv)
             m = blk.m
-            L = torch.linalg.cholesky(sigma)
+            L = torch.linalg.cholesky(sigma)  # (D, m, m)
             W = torch.linalg.solve_triangular(
-                L, blk.RHS[:m, :], upper=False
-            )
-            wu = W[:, -1]
-            Ws, Wc = W[:, 0:-1:2], W[:, 1:-1:2]
-            out[i, 0] = blk.sNs[0] - (Ws * Ws).sum(0)
-            out[i, 1] = blk.sNs[1] - (Wc * Wc).sum(0)
-            out[i, 2] = blk.sNs[2] - (Ws * Wc).sum(0)
-            out[i, 3] = blk.sNr[0] - Ws.transpose(0, 1) @ wu
-            out[i, 4] = blk.sNr[1] - Wc.transpose(0, 1) @ wu
-        return out
+                L, blk.RHS[:m, :][None].expand(D, -1, -1), upper=False
+            )  # (D, m, 2F+1)
+            wu = W[:, :, -1]
+            Ws, Wc = W[:, :, 0:-1:2], W[:, :, 1:-1:2]
+            out[i, :, 0] = blk.sNs[0][None] - (Ws * Ws).sum(1)
+            out[i, :, 1] = blk.sNs[1][None] - (Wc * Wc).sum(1)
+            out[i, :, 2] = blk.sNs[2][None] - (Ws * Wc).sum(1)
+            out[i, :, 3] = blk.sNr[0][None] - torch.einsum(
+                "dmf,dm->df", Ws, wu)
+            out[i, :, 4] = blk.sNr[1][None] - torch.einsum(
+                "dmf,dm->df", Wc, wu)
+        return out[:, 0] if not batched else out
 
     def _accum_eager(self, blk: PulsarBlock, sigma, fp_out):
         """Eager per-pulsar accumulation: Cholesky + TRSM + fused 2x2."""

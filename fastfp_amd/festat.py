@@ -196,3 +196,89 @@ def compute_Fe(psrs, pta, noise, freqs, sky, device=None) -> np.ndarray:
     Nvecs, Ts, sigmas = get_mats_fp(pta, noise)
     return FastFe(psrs, pta).sweep(freqs, sky, Nvecs, Ts, sigmas,
                                    device=device)
+
+
+def _assemble_fe_draws(prods, fplus, fcross, rcond=1e-12):
+    """Draw-batched Fe assembly: ``prods`` (P, D, 5, F) -> (D, F)."""
+    ss, cc, sc = prods[:, :, 0], prods[:, :, 1], prods[:, :, 2]
+    sr, cr = prods[:, :, 3], prods[:, :, 4]
+    fp2, fx2, fpx = fplus**2, fcross**2, fplus * fcross
+
+    def w(coeff, q):  # (P,) x (P, D, F) -> (D, F)
+        return np.einsum("p,pdf->df", coeff, q)
+
+    D, F = prods.shape[1], prods.shape[3]
+    M = np.empty((D, F, 4, 4))
+    M[..., 0, 0] = w(fp2, ss)
+    M[..., 0, 1] = M[..., 1, 0] = w(fp2, sc)
+    M[..., 0, 2] = M[..., 2, 0] = w(fpx, ss)
+    M[..., 0, 3] = M[..., 3, 0] = w(fpx, sc)
+    M[..., 1, 1] = w(fp2, cc)
+    M[..., 1, 2] = M[..., 2, 1] = w(fpx, sc)
+    M[..., 1, 3] = M[..., 3, 1] = w(fpx, cc)
+    M[..., 2, 2] = w(fx2, ss)
+    M[..., 2, 3] = M[..., 3, 2] = w(fx2, sc)
+    M[..., 3, 3] = w(fx2, cc)
+    N = np.stack(
+        [w(fplus, sr), w(fplus, cr), w(fcross, sr), w(fcross, cr)], axis=-1
+    )  # (D, F, 4)
+    try:
+        x = np.linalg.solve(M, N[..., None])[..., 0]
+    except np.linalg.LinAlgError:
+        flatM = M.reshape(-1, 4, 4)
+        flatN = N.reshape(-1, 4)
+        x = np.stack([np.linalg.pinv(Mi, rcond=rcond) @ Ni
+                      for Mi, Ni in zip(flatM, flatN)]).reshape(D, F, 4)
+    return 0.5 * np.einsum("dfi,dfi->df", N, x)
+
+
+class NMFe:
+    """Noise-marginalized Fe: the Fe statistic evaluated across MCMC
+    red-noise draws, symmetric to :class:`fastfp_amd.NMFp` for the
+    sky-coherent statistic (no reference counterpart — the reference
+    has neither Fe nor its marginalized form).
+
+    ``rn_sigs``: the per-pulsar phi containers (``pta.rn_containers``).
+    """
+
+    def __init__(self, psrs, rn_sigs):
+        self.psrs = psrs
+        self.rn_sigs = rn_sigs
+        self.fe = FastFe(psrs)  # validates .pos, holds antenna inputs
+
+    def sweep(
+        self,
+        freqs,
+        sky,
+        samples: dict,
+        Nvecs,
+        Ts,
+        device: str = None,
+        draw_chunk: int = 128,
+        freq_chunk: int = 2048,
+        engine: FpEngine = None,
+    ) -> np.ndarray:
+        """Fe over (draws x sky x freqs).  ``samples``: parameter-name
+        -> (D,) arrays (the ``map_params`` format).  Returns
+        (D, nsky, F)."""
+        from fastfp_amd.noise import batch_phiinv
+
+        if device is None:
+            device = "cuda" if torch.cuda.is_available() else "cpu"
+        if engine is None:
+            engine = FpEngine(self.psrs, Nvecs, Ts, device=device)
+            engine.precompute(freqs, freq_chunk=freq_chunk)
+        phiinvs = batch_phiinv(self.rn_sigs, samples)
+        phiinvs = [p[None, :] if p.dim() == 1 else p for p in phiinvs]
+        D = phiinvs[0].shape[0]
+        F = engine.freqs.shape[0]
+        ant = [gw_antenna_pattern(self.fe.pos, th, ph) for th, ph in sky]
+        out = np.empty((D, len(sky), F))
+        for lo in range(0, D, draw_chunk):
+            hi = min(lo + draw_chunk, D)
+            prods = engine.sweep_products(
+                phiinvs=[p[lo:hi] for p in phiinvs]
+            ).cpu().numpy()  # (P, Dc, 5, F)
+            for k, (fplus, fcross) in enumerate(ant):
+                out[lo:hi, k, :] = _assemble_fe_draws(prods, fplus, fcross)
+        return out

@@ -231,3 +231,70 @@ def test_fe_sweep_gpu_matches_cpu():
     g = fe.sweep(freqs, sky, Nvecs, Ts, sigmas, device="cuda:0")
     c = fe.sweep(freqs, sky, Nvecs, Ts, sigmas, device="cpu")
     np.testing.assert_allclose(g, c, rtol=1e-9)
+
+
+# ----------------------------------------------------------------------
+# noise-marginalized Fe
+# ----------------------------------------------------------------------
+def test_nmfe_matches_per_draw_fe():
+    """NMFe.sweep (draw-batched products) equals FastFe evaluated at
+    each draw's fixed noise, across draw-chunk boundaries."""
+    from fastfp_amd.festat import NMFe
+    from fastfp_amd.model import get_mats_nmfp
+
+    psrs, noise, pta = _pta(npsr=3, ntoa=80, seed=17)
+    TNTs, Nvecs, Ts = get_mats_nmfp(pta, noise)
+    D = 5
+    rng = np.random.default_rng(8)
+    samples = {
+        n: (rng.uniform(2, 6, D) if n.endswith("gamma")
+            else rng.uniform(-16, -14, D))
+        for n in pta.params
+    }
+    freqs = np.array([6e-9, 1.7e-8])
+    sky = [(0.9, 1.1), (2.2, 4.0)]
+    nm = NMFe(psrs, pta.rn_containers)
+    got = nm.sweep(freqs, sky, samples, Nvecs, Ts, device="cpu",
+                   draw_chunk=2)
+    assert got.shape == (D, 2, 2)
+
+    fe = FastFe(psrs)
+    for d in range(D):
+        pt = {k: v[d] for k, v in samples.items()}
+        sigmas = [
+            np.asarray(TNT) + np.diag(c.get_phiinv(pt).numpy())
+            for TNT, c in zip(TNTs, pta.rn_containers)
+        ]
+        for k, (th, ph) in enumerate(sky):
+            for fi, f in enumerate(freqs):
+                want = fe.calculate_Fe(f, th, ph, Nvecs, Ts, sigmas)
+                np.testing.assert_allclose(got[d, k, fi], want, rtol=1e-8,
+                                           err_msg=f"d={d} k={k} f={fi}")
+
+
+def test_sweep_products_batched_matches_fixed():
+    """(D, m) phiinvs products equal D stacked fixed-noise calls."""
+    from fastfp_amd import FpEngine
+    from fastfp_amd.model import get_mats_nmfp
+    from fastfp_amd.noise import batch_phiinv
+    import torch
+
+    psrs, noise, pta = _pta(npsr=2, ntoa=60, seed=19)
+    _, Nvecs, Ts = get_mats_nmfp(pta, noise)
+    D = 3
+    rng = np.random.default_rng(4)
+    samples = {
+        n: torch.as_tensor(
+            rng.uniform(2, 6, D) if n.endswith("gamma")
+            else rng.uniform(-16, -14, D), dtype=torch.float64)
+        for n in pta.params
+    }
+    eng = FpEngine(psrs, Nvecs, Ts, device="cpu")
+    eng.precompute(np.array([7e-9, 2e-8, 3e-8]))
+    piv = batch_phiinv(pta.rn_containers, samples)
+    piv = [p[None, :] if p.dim() == 1 else p for p in piv]
+    batched = eng.sweep_products(phiinvs=piv).numpy()  # (P, D, 5, F)
+    for d in range(D):
+        fixed = eng.sweep_products(
+            phiinvs=[p[d] for p in piv]).numpy()  # (P, 5, F)
+        np.testing.assert_array_equal(batched[:, d], fixed)
