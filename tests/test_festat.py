@@ -298,3 +298,27 @@ def test_sweep_products_batched_matches_fixed():
         fixed = eng.sweep_products(
             phiinvs=[p[d] for p in piv]).numpy()  # (P, 5, F)
         np.testing.assert_array_equal(batched[:, d], fixed)
+
+
+def test_fe_scaling_and_positivity_random_configs():
+    """Properties across random configs/skies: Fe >= 0 (maximized
+    quadratic form over the filter span) and Fe(lambda * r) =
+    lambda^2 * Fe(r) (N scales linearly, M is residual-independent)."""
+    for seed in range(6):
+        rng = np.random.default_rng(100 + seed)
+        psrs, noise, pta = _pta(npsr=int(rng.integers(2, 5)),
+                                ntoa=int(rng.integers(50, 110)),
+                                seed=200 + seed)
+        Nvecs, Ts, sigmas = get_mats_fp(pta, noise)
+        freqs = np.sort(rng.uniform(5e-9, 5e-8, 3))
+        sky = [(float(rng.uniform(0.2, np.pi - 0.2)),
+                float(rng.uniform(0, 2 * np.pi)))]
+        fe = FastFe(psrs, pta)
+        base = fe.sweep(freqs, sky, Nvecs, Ts, sigmas, device="cpu")
+        assert (base >= -1e-12).all(), base
+        lam = 3.0
+        for p in psrs:
+            p.residuals = p.residuals * lam
+        fe2 = FastFe(psrs, pta)
+        scaled = fe2.sweep(freqs, sky, Nvecs, Ts, sigmas, device="cpu")
+        np.testing.assert_allclose(scaled, lam**2 * base, rtol=1e-9)
