@@ -209,3 +209,20 @@ def test_fibonacci_sky_coverage():
     # equal-area: z = cos(theta) should be ~uniform on [-1, 1]
     z = np.sort(np.cos(th))
     assert np.abs(z - np.linspace(z[0], z[-1], 100)).max() < 0.03
+
+
+def test_run_nmfe_output(inputs, tmp_path):
+    """NM-Fe CLI: (nsamples, nsky, nfreqs) npy + axes metadata."""
+    from fastfp_amd.cli import run_nmfe
+
+    tmp, psrfile, noisefile, chainfile = inputs
+    run_nmfe.main(psrfile, noisefile, chainfile, "nmfe", inc_cp=True,
+                  nrncomps=3, ngwbcomps=3, ncwfreqs=4, nsamples=5,
+                  nsky=3, outdir=str(tmp_path), device="cpu",
+                  batch_size=2)
+    out = np.load(tmp_path / "nmfe.npy")
+    assert out.shape == (5, 3, 4)
+    assert np.isfinite(out).all()
+    with open(tmp_path / "nmfe.meta.json") as f:
+        meta = json.load(f)
+    assert len(meta["freqs"]) == 4 and len(meta["sky"]) == 3
