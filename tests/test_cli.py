@@ -226,3 +226,26 @@ def test_run_nmfe_output(inputs, tmp_path):
     with open(tmp_path / "nmfe.meta.json") as f:
         meta = json.load(f)
     assert len(meta["freqs"]) == 4 and len(meta["sky"]) == 3
+
+
+def test_run_nmfe_torchrun_ws2_gloo(inputs, tmp_path):
+    """NM-Fe CLI under the multi-rank launch (2 ranks, gloo): draw
+    sharding + gather across the (draws, sky, freqs) output."""
+    import subprocess
+    import sys
+
+    tmp, psrfile, noisefile, chainfile = inputs
+    repo = os.path.abspath(os.path.join(os.path.dirname(__file__), ".."))
+    r = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+         "--master-port", "29884", "-m", "fastfp_amd.cli.run_nmfe",
+         psrfile, noisefile, chainfile, "nmfe2", "--inc_cp",
+         "--nrncomps", "3", "--ngwbcomps", "3", "--ncwfreqs", "3",
+         "--nsamples", "5", "--nsky", "2", "--outdir", str(tmp_path),
+         "--device", "cpu", "--batch_size", "2"],
+        cwd=repo, capture_output=True, text=True, timeout=600,
+    )
+    assert r.returncode == 0, r.stderr[-2000:]
+    out = np.load(tmp_path / "nmfe2.npy")
+    assert out.shape == (5, 2, 3) and np.isfinite(out).all()
