@@ -322,3 +322,33 @@ def test_fe_scaling_and_positivity_random_configs():
         fe2 = FastFe(psrs, pta)
         scaled = fe2.sweep(freqs, sky, Nvecs, Ts, sigmas, device="cpu")
         np.testing.assert_allclose(scaled, lam**2 * base, rtol=1e-9)
+
+
+@pytest.mark.gpu
+def test_nmfe_sweep_gpu_matches_cpu():
+    """Draw-batched NM-Fe on the HIP engine vs CPU — validated on
+    MI355X (profiles/r02_nmfe_gpu_validation.log: rel err 8e-16)."""
+    import torch
+
+    from fastfp_amd.festat import NMFe
+    from fastfp_amd.model import get_mats_nmfp
+
+    if not torch.cuda.is_available():
+        pytest.skip("needs a GPU")
+    psrs, noise, pta = _pta(npsr=2, ntoa=120, seed=6)
+    TNTs, Nvecs, Ts = get_mats_nmfp(pta, noise)
+    rng = np.random.default_rng(2)
+    D = 4
+    samples = {
+        n: (rng.uniform(2, 6, D) if n.endswith("gamma")
+            else rng.uniform(-16, -14, D))
+        for n in pta.params
+    }
+    freqs = np.array([7e-9, 2e-8])
+    sky = [(0.9, 1.4)]
+    nm = NMFe(psrs, pta.rn_containers)
+    g = nm.sweep(freqs, sky, samples, Nvecs, Ts, device="cuda:0",
+                 draw_chunk=2)
+    c = nm.sweep(freqs, sky, samples, Nvecs, Ts, device="cpu",
+                 draw_chunk=2)
+    np.testing.assert_allclose(g, c, rtol=1e-9)
