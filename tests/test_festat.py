@@ -352,3 +352,26 @@ def test_nmfe_sweep_gpu_matches_cpu():
     c = nm.sweep(freqs, sky, samples, Nvecs, Ts, device="cpu",
                  draw_chunk=2)
     np.testing.assert_allclose(g, c, rtol=1e-9)
+
+
+def test_fe_with_block_noise_ecorr_kernel():
+    """Fe on an EcorrKernelNoise model (block-diagonal N): the engine
+    products path and the get_xCy BlockNoise parity branch agree."""
+    psrs = make_synthetic_pta(npsr=3, ntoa=90, ntm=3, seed=23)
+    noise = {"gw_gamma": 13.0 / 3.0, "gw_log10_A": float(np.log10(2e-15))}
+    for p in psrs:
+        noise[f"{p.name}_red_noise_gamma"] = 4.0
+        noise[f"{p.name}_red_noise_log10_A"] = -14.5
+        for b in np.unique(p.backend_flags):
+            noise[f"{p.name}_basis_ecorr_{b}_log10_ecorr"] = -6.5
+    pta = initialize_pta(psrs, noise, inc_cp=True, rn_comps=3,
+                         gwb_comps=3, ecorr_kernel=True)
+    Nvecs, Ts, sigmas = get_mats_fp(pta, noise)
+    freqs = np.array([8e-9, 2.2e-8])
+    sky = [(1.2, 0.6)]
+    fe = FastFe(psrs, pta)
+    grid = fe.sweep(freqs, sky, Nvecs, Ts, sigmas, device="cpu")
+    assert np.isfinite(grid).all()
+    for fi, f in enumerate(freqs):
+        want = fe.calculate_Fe(f, sky[0][0], sky[0][1], Nvecs, Ts, sigmas)
+        np.testing.assert_allclose(grid[0, fi], want, rtol=1e-7)
