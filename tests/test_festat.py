@@ -375,3 +375,43 @@ def test_fe_with_block_noise_ecorr_kernel():
     for fi, f in enumerate(freqs):
         want = fe.calculate_Fe(f, sky[0][0], sky[0][1], Nvecs, Ts, sigmas)
         np.testing.assert_allclose(grid[0, fi], want, rtol=1e-7)
+
+
+def test_nmfe_with_gp_ecorr_model():
+    """NM-Fe on a GP-ECORR model (heterogeneous phi containers: fixed
+    ecorr block + sampled rn bins) vs per-draw FastFe."""
+    from fastfp_amd.festat import NMFe
+    from fastfp_amd.model import get_mats_nmfp
+
+    psrs = make_synthetic_pta(npsr=2, ntoa=80, ntm=3, seed=29)
+    noise = {"gw_gamma": 13.0 / 3.0, "gw_log10_A": float(np.log10(2e-15))}
+    for p in psrs:
+        noise[f"{p.name}_red_noise_gamma"] = 4.0
+        noise[f"{p.name}_red_noise_log10_A"] = -14.5
+        for b in np.unique(p.backend_flags):
+            noise[f"{p.name}_basis_ecorr_{b}_log10_ecorr"] = -6.5
+    pta = initialize_pta(psrs, noise, inc_cp=True, rn_comps=3,
+                         gwb_comps=2, inc_ecorr=True)
+    TNTs, Nvecs, Ts = get_mats_nmfp(pta, noise)
+    rng = np.random.default_rng(6)
+    D = 3
+    samples = {
+        n: (rng.uniform(2, 6, D) if n.endswith("gamma")
+            else rng.uniform(-16, -14, D))
+        for n in pta.params
+    }
+    freqs = np.array([9e-9, 2.4e-8])
+    sky = [(0.8, 3.3)]
+    nm = NMFe(psrs, pta.rn_containers)
+    got = nm.sweep(freqs, sky, samples, Nvecs, Ts, device="cpu")
+    fe = FastFe(psrs)
+    for d in range(D):
+        pt = {k: v[d] for k, v in samples.items()}
+        sigmas = [
+            np.asarray(TNT) + np.diag(c.get_phiinv(pt).numpy())
+            for TNT, c in zip(TNTs, pta.rn_containers)
+        ]
+        for fi, f in enumerate(freqs):
+            want = fe.calculate_Fe(f, sky[0][0], sky[0][1],
+                                   Nvecs, Ts, sigmas)
+            np.testing.assert_allclose(got[d, 0, fi], want, rtol=1e-7)
