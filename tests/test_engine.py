@@ -612,3 +612,38 @@ def test_graph_cache_keyed_by_engine_identity():
     key_b = (id(eng_b), 3, 4, tuple(samples))
     assert key_b != key_a
     assert nm._graph_seen[key_b] == 1 and key_b not in nm._graphs
+
+
+def test_duplicate_toas_all_noise_modes():
+    """EXACT duplicate TOAs (simultaneous multi-channel observations,
+    common in real backends): engine == parity in all three noise
+    modes (plain / GP-ECORR / EcorrKernelNoise)."""
+    from fastfp_amd.data import PulsarData
+
+    rng = np.random.default_rng(0)
+    n = 60
+    base = np.sort(rng.uniform(0, 4e8, n // 3))
+    toas = np.repeat(base, 3)
+    M = np.stack([np.ones(n), toas / toas.max(),
+                  (toas / toas.max()) ** 2], axis=1)
+    psr = PulsarData(
+        name="JDUP", toas=toas, toaerrs=np.full(n, 1e-6),
+        residuals=rng.normal(0, 1e-6, n), Mmat=M,
+        backend_flags=np.array(["A"] * n, dtype=object),
+        pos=np.array([0.3, 0.5, 0.8]),
+    )
+    noise = {"gw_gamma": 13 / 3, "gw_log10_A": float(np.log10(2e-15)),
+             "JDUP_red_noise_gamma": 4.0,
+             "JDUP_red_noise_log10_A": -14.5,
+             "JDUP_basis_ecorr_A_log10_ecorr": -6.5}
+    for ek, ie in ((False, False), (False, True), (True, False)):
+        pta = initialize_pta([psr], noise, inc_cp=True, rn_comps=3,
+                             gwb_comps=2, ecorr_kernel=ek, inc_ecorr=ie)
+        Nvecs, Ts, sigmas = get_mats_fp(pta, noise)
+        freqs = np.array([8e-9, 2e-8])
+        fp_obj = FastFp([psr])
+        want = np.array([fp_obj.calculate_Fp(f, Nvecs, Ts, sigmas)
+                         for f in freqs])
+        got = fp_obj.sweep(freqs, Nvecs, Ts, sigmas, device="cpu")
+        np.testing.assert_allclose(got, want, rtol=1e-6,
+                                   err_msg=f"ek={ek} ie={ie}")
