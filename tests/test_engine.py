@@ -647,3 +647,43 @@ def test_duplicate_toas_all_noise_modes():
         got = fp_obj.sweep(freqs, Nvecs, Ts, sigmas, device="cpu")
         np.testing.assert_allclose(got, want, rtol=1e-6,
                                    err_msg=f"ek={ek} ie={ie}")
+
+
+def test_degenerate_full_span_timing_model_nonfinite_consistently():
+    """A timing model whose basis spans the ENTIRE residual space
+    absorbs every filter (M -> 0, N -> 0): Fp is mathematically
+    undefined (0/0).  Both paths must agree on non-finite output (the
+    CLIs warn on non-finite values; bench asserts finiteness on
+    healthy models) — this documents the behavior rather than
+    silently guessing a value."""
+    psr = make_synthetic_pta(npsr=1, ntoa=12, ntm=3, seed=2,
+                             ragged=False)[0]
+    rng = np.random.default_rng(1)
+    psr.Mmat = np.linalg.qr(rng.normal(size=(12, 12)))[0]  # complete basis
+    noise = {"gw_gamma": 13 / 3, "gw_log10_A": float(np.log10(2e-15)),
+             f"{psr.name}_red_noise_gamma": 4.0,
+             f"{psr.name}_red_noise_log10_A": -14.5}
+    pta = initialize_pta([psr], noise, inc_cp=True, rn_comps=2,
+                         gwb_comps=2)
+    Nvecs, Ts, sigmas = get_mats_fp(pta, noise)
+    got = FastFp([psr]).sweep(np.array([1e-8]), Nvecs, Ts, sigmas,
+                              device="cpu")
+    want = FastFp([psr]).calculate_Fp(1e-8, Nvecs, Ts, sigmas)
+    assert not np.isfinite(got).all()
+    assert not np.isfinite(want)
+
+
+def test_zero_residuals_give_zero_fp():
+    """No signal, no noise realization -> Fp exactly 0 (N = 0)."""
+    psrs = make_synthetic_pta(npsr=2, ntoa=60, ntm=3, seed=1)
+    for p in psrs:
+        p.residuals = np.zeros_like(p.residuals)
+    noise = {"gw_gamma": 13 / 3, "gw_log10_A": float(np.log10(2e-15))}
+    for p in psrs:
+        noise[f"{p.name}_red_noise_gamma"] = 4.0
+        noise[f"{p.name}_red_noise_log10_A"] = -14.5
+    pta = initialize_pta(psrs, noise, inc_cp=True, rn_comps=3, gwb_comps=2)
+    Nvecs, Ts, sigmas = get_mats_fp(pta, noise)
+    got = FastFp(psrs).sweep(np.array([1e-8, 3e-8]), Nvecs, Ts, sigmas,
+                             device="cpu")
+    np.testing.assert_array_equal(got, 0.0)
