@@ -41,3 +41,4 @@ def test_walkthrough_notebook_code_runs(tmp_path, monkeypatch):
     out = buf.getvalue()
     assert "KS test vs chi2(24)" in out
     assert "vals shape: (32, 100)" in out
+    assert "Fe map shape: (6, 200)" in out
