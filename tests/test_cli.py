@@ -249,3 +249,27 @@ def test_run_nmfe_torchrun_ws2_gloo(inputs, tmp_path):
     assert r.returncode == 0, r.stderr[-2000:]
     out = np.load(tmp_path / "nmfe2.npy")
     assert out.shape == (5, 2, 3) and np.isfinite(out).all()
+
+
+def test_run_nmfp_more_ranks_than_draws(inputs, tmp_path):
+    """world_size 3 with only 2 draws: the zero-draw rank must
+    participate in the padded all-gather and the output stays
+    (nsamples, nfreqs)."""
+    import subprocess
+    import sys
+
+    tmp, psrfile, noisefile, chainfile = inputs
+    repo = os.path.abspath(os.path.join(os.path.dirname(__file__), ".."))
+    r = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "3", "--master-addr", "127.0.0.1",
+         "--master-port", "29893", "-m", "fastfp_amd.cli.run_nmfp",
+         psrfile, noisefile, chainfile, "zd", "--inc_cp",
+         "--nrncomps", "3", "--ngwbcomps", "3", "--ncwfreqs", "3",
+         "--nsamples", "2", "--outdir", str(tmp_path),
+         "--device", "cpu"],
+        cwd=repo, capture_output=True, text=True, timeout=600,
+    )
+    assert r.returncode == 0, r.stderr[-2000:]
+    out = np.load(tmp_path / "zd.npy")
+    assert out.shape == (2, 3) and np.isfinite(out).all()
